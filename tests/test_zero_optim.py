@@ -1,0 +1,98 @@
+"""Bf16ZeroOptimizer differential test vs DDP+Adam (gloo world_size=2, CPU).
+
+Mirrors the reference's examples/test_zero_optim.py oracle method.
+"""
+
+import copy
+
+import torch
+import torch.nn as nn
+
+from tests.dist_helpers import run_distributed
+
+
+def _make_model(seed=0):
+    torch.manual_seed(seed)
+    return nn.Sequential(nn.Linear(48, 96), nn.Tanh(), nn.Linear(96, 48),
+                         nn.Tanh(), nn.Linear(48, 8))
+
+
+def _zero_vs_ddp(rank, world_size, stage2=False):
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+
+    model_a = _make_model(seed=11)
+    model_b = copy.deepcopy(model_a)
+
+    inner = torch.optim.Adam(model_a.parameters(), lr=1e-3)
+    zopt = Bf16ZeroOptimizer(inner, stage2=stage2)
+
+    ref = nn.parallel.DistributedDataParallel(model_b)
+    opt_b = torch.optim.Adam(model_b.parameters(), lr=1e-3)
+
+    for it in range(5):
+        torch.manual_seed(500 + 10 * it + rank)
+        x = torch.randn(6, 48)
+
+        model_a(x).pow(2).mean().backward()
+        ref(x).pow(2).mean().backward()
+
+        zopt.step()
+        opt_b.step()
+        zopt.zero_grad()
+        opt_b.zero_grad()
+
+        for (na, pa), (nb, pb) in zip(model_a.named_parameters(),
+                                      model_b.named_parameters()):
+            assert torch.allclose(pa, pb, atol=1e-5), \
+                f"iter {it} param {na}: max diff " \
+                f"{(pa - pb).abs().max().item()}"
+    return True
+
+
+def test_zero1():
+    run_distributed(_zero_vs_ddp, world_size=2)
+
+
+def test_zero2():
+    run_distributed(_zero_vs_ddp, world_size=2, kwargs={"stage2": True})
+
+
+def _zero_single_rank(rank, world_size):
+    # world_size=1 path: master-grad copy without comm
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+
+    model = _make_model(seed=2)
+    ref = copy.deepcopy(model)
+    zopt = Bf16ZeroOptimizer(torch.optim.Adam(model.parameters(), lr=1e-3))
+    opt_ref = torch.optim.Adam(ref.parameters(), lr=1e-3)
+    for it in range(3):
+        torch.manual_seed(it)
+        x = torch.randn(4, 48)
+        model(x).sum().backward()
+        ref(x).sum().backward()
+        zopt.step()
+        opt_ref.step()
+        zopt.zero_grad()
+        opt_ref.zero_grad()
+    for pa, pb in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6)
+    return True
+
+
+def test_zero_world1():
+    run_distributed(_zero_single_rank, world_size=1)
+
+
+def _zero_clip(rank, world_size):
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+    model = _make_model(seed=5)
+    zopt = Bf16ZeroOptimizer(torch.optim.Adam(model.parameters(), lr=1e-3),
+                             clip_grad=0.1)
+    torch.manual_seed(rank)
+    (model(torch.randn(4, 48) * 100).pow(2).mean()).backward()
+    zopt.step()  # just verifies the clip path runs distributed
+    return True
+
+
+def test_zero_clip_grad():
+    run_distributed(_zero_clip, world_size=2)

@@ -1,0 +1,157 @@
+"""MoE-DP: gradient sync for replicated experts across the moe_dp groups.
+
+Capability parity with the reference MoEDP + hooks API
+(/root/reference/torchdistpackage/ddp/naive_ddp.py:233-441 and ddp/moe_dp.md):
+expert params living under expert parallelism are *replicated* across the
+moe_dp group (ranks holding the same expert shard); their grads all-reduce over
+'moe_dp' instead of 'data'.  Provides the module-level ``moe_dp_iter_step()``
+and ``create_moe_dp_hooks(...)`` factory the reference exports.
+
+Fixes vs the reference: the undefined-variable bug in its sync path
+(naive_ddp.py:401) and the never-selected "sum" reduce-op
+(``reduce_op.lower`` missing parens, :53) are not replicated; AVG is the only
+mode here (SUM+div on gloo).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Iterable, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from .naive_ddp import GradBucket, _align
+
+
+class MoEDP:
+    """Bucketed grad all-reduce over the moe_dp group for a set of expert
+    params (given as an iterable or dict name->param)."""
+
+    def __init__(self, params, group: Optional[dist.ProcessGroup] = None,
+                 broadcast_src: bool = True, bucket_cap_mb: float = 50.0,
+                 num_grad_acc_iter: int = 1):
+        if isinstance(params, dict):
+            self._params = [p for p in params.values() if p.requires_grad]
+        else:
+            self._params = [p for p in params if p.requires_grad]
+        if group is None:
+            from ..dist.topo import tpc
+            group = tpc.get_group("moe_dp")
+        self.group = group
+        self.num_grad_acc_iter = max(1, num_grad_acc_iter)
+        self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
+        self._fires: Dict[int, int] = {}
+        self._use_gpu = torch.cuda.is_available() and \
+            any(p.is_cuda for p in self._params)
+        self._reduce_stream = torch.cuda.Stream() if self._use_gpu else None
+        self._pending = []
+        self._works = []
+        self._hooks = []
+
+        if broadcast_src and dist.is_initialized() and self._world() > 1:
+            src = dist.get_process_group_ranks(self.group)[0]
+            with torch.no_grad():
+                for p in self._params:
+                    dist.broadcast(p.data, src=src, group=self.group)
+
+        self._buckets: List[GradBucket] = []
+        self._param_bucket: Dict[int, tuple] = {}
+        self._build_buckets()
+        for p in self._params:
+            self._hooks.append(
+                p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    def _world(self) -> int:
+        return dist.get_world_size(self.group) if dist.is_initialized() else 1
+
+    def _build_buckets(self):
+        cur, cur_bytes = [], 0
+        for p in reversed(self._params):
+            nbytes = _align(p.numel()) * p.element_size()
+            if cur and cur_bytes + nbytes > self.bucket_cap:
+                self._close(cur)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += nbytes
+        if cur:
+            self._close(cur)
+
+    def _close(self, params):
+        b = GradBucket(params, params[0].dtype, params[0].device)
+        for i, p in enumerate(params):
+            self._param_bucket[id(p)] = (b, i)
+        self._buckets.append(b)
+
+    def _on_grad_ready(self, p: torch.Tensor):
+        if self._world() == 1:
+            return
+        fires = self._fires.get(id(p), 0) + 1
+        self._fires[id(p)] = fires
+        if fires % self.num_grad_acc_iter != 0:
+            return
+        bucket, idx = self._param_bucket[id(p)]
+        if bucket.push(idx, p.grad):
+            if self._use_gpu:
+                cur = torch.cuda.current_stream()
+                ev = torch.cuda.Event()
+                ev.record(cur)
+                with torch.cuda.stream(self._reduce_stream):
+                    self._reduce_stream.wait_event(ev)
+                    dist.all_reduce(bucket.data, op=dist.ReduceOp.AVG,
+                                    group=self.group)
+                    done = torch.cuda.Event()
+                    done.record(self._reduce_stream)
+                    self._pending.append(done)
+            else:
+                w = dist.all_reduce(bucket.data, op=dist.ReduceOp.SUM,
+                                    group=self.group, async_op=True)
+                self._works.append((w, bucket))
+            bucket.reset()
+
+    @torch.no_grad()
+    def reduce_gradients(self):
+        if self._world() == 1:
+            return
+        if self._use_gpu:
+            cur = torch.cuda.current_stream()
+            for ev in self._pending:
+                cur.wait_event(ev)
+            self._pending.clear()
+            for b in self._buckets:
+                if b.ready > 0:
+                    dist.all_reduce(b.data, op=dist.ReduceOp.AVG,
+                                    group=self.group)
+                    b.reset()
+        else:
+            for w, b in self._works:
+                w.wait()
+                b.data.div_(self._world())
+            self._works.clear()
+            for b in self._buckets:
+                if b.ready > 0:
+                    dist.all_reduce(b.data, op=dist.ReduceOp.SUM,
+                                    group=self.group)
+                    b.data.div_(self._world())
+                    b.reset()
+        for b in self._buckets:
+            for p, v in zip(b.params, b.views):
+                if p.grad is not None:
+                    p.grad.copy_(v)
+
+
+# module-level convenience API, reference-compatible
+# (naive_ddp.py:414-441)
+_moe_dp_mod: Optional[MoEDP] = None
+
+
+def create_moe_dp_hooks(params, group: Optional[dist.ProcessGroup] = None,
+                        broadcast_src: bool = True, **kw) -> MoEDP:
+    global _moe_dp_mod
+    _moe_dp_mod = MoEDP(params, group=group, broadcast_src=broadcast_src, **kw)
+    return _moe_dp_mod
+
+
+def moe_dp_iter_step():
+    """Call once per iteration (before optimizer.step())."""
+    if _moe_dp_mod is not None:
+        _moe_dp_mod.reduce_gradients()
