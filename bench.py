@@ -1,0 +1,181 @@
+"""Flagship benchmark: GPT-2 1.3B training step, tokens/sec whole-node.
+
+BASELINE.json metric: "tokens/sec (whole node) GPT-2-1.3B DP+TP2+PP2 at
+1/2/4/8 MI355X".  Parallelism by GPU count (dist_config ordered
+[data, pipe, tensor], tensor innermost):
+
+    N=1: plain single GPU          N=2: [('data',1),('pipe',1),('tensor',2)]
+    N=4: [('data',2),('tensor',2)] N=8: [('data',2),('pipe',2),('tensor',2)]
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+Prints ONE JSON line from rank 0 (whole-job aggregate tokens/s).
+Synthetic data (random token ids), random-init weights, bf16 compute,
+fp32 master weights + fused HIP AdamW.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=8,
+                    help="micro/global batch per DP rank")
+    ap.add_argument("--seq", type=int, default=1024)
+    ap.add_argument("--model", type=str, default="gpt2_1.3b",
+                    choices=["gpt2_1.3b", "gpt2_small", "tiny"])
+    ap.add_argument("--micro-batches", type=int, default=8,
+                    help="micro-batches per step when PP is active")
+    return ap.parse_args()
+
+
+PARALLEL_MAP = {
+    1: (1, 1, 1),
+    2: (1, 1, 2),
+    4: (2, 1, 2),
+    8: (2, 2, 2),
+}
+
+
+def main():
+    args = parse_args()
+    from torchdistpackage_amd import setup_distributed, tpc, fix_rand
+    from torchdistpackage_amd.ddp import NaiveDdp
+    from torchdistpackage_amd.models.gpt2 import (GPT2Config, GPT2Model,
+                                                  gpt2_small, gpt2_xl_1p3b)
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    n_gpus = args.gpus if args.gpus > 1 else world
+    if world > 1:
+        info = setup_distributed()
+        rank = info["rank"]
+    else:
+        rank = 0
+        if torch.cuda.is_available():
+            torch.cuda.set_device(0)
+    import torch.distributed as dist
+
+    dp, pp, tp = PARALLEL_MAP.get(world, (world, 1, 1))
+    if world > 1:
+        tpc.setup_process_groups(
+            [("data", dp), ("pipe", pp), ("tensor", tp)])
+    use_pp = pp > 1
+
+    fix_rand(tpc.get_dp_rank() if world > 1 else 0)
+
+    if args.model == "gpt2_1.3b":
+        cfg = gpt2_xl_1p3b()
+    elif args.model == "gpt2_small":
+        cfg = gpt2_small()
+    else:
+        cfg = GPT2Config(vocab_size=2048, n_layer=4, n_head=8, dim=512,
+                         max_seq=args.seq)
+    cfg.max_seq = max(cfg.max_seq, args.seq)
+
+    dev = torch.device("cuda", torch.cuda.current_device()) \
+        if torch.cuda.is_available() else torch.device("cpu")
+    dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
+
+    if use_pp:
+        from bench_pp import run_pp_bench  # PP path in separate module
+        result = run_pp_bench(args, cfg, dev, dtype, dp, pp, tp)
+    else:
+        result = run_dp_tp_bench(args, cfg, dev, dtype, dp, tp)
+
+    if rank == 0:
+        B_global = args.batch * dp * args.micro_batches if use_pp \
+            else args.batch * dp
+        tokens_per_step = B_global * args.seq
+        toks_per_s = tokens_per_step / (result["ms_per_step"] / 1e3)
+        print(json.dumps({
+            "metric": "tokens_per_second",
+            "value": toks_per_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": result["ms_per_step"],
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": B_global,
+                "seq_len": args.seq,
+                "parallelism": f"dp{dp}_pp{pp}_tp{tp}",
+            },
+        }), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+
+
+def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
+    import torch.distributed as dist
+    from torchdistpackage_amd import tpc
+    from torchdistpackage_amd.ddp import NaiveDdp
+    from torchdistpackage_amd.models.gpt2 import GPT2Model
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    torch.manual_seed(1234)  # same init across ranks (then broadcast anyway)
+    model = GPT2Model(cfg, device=dev, dtype=dtype)
+
+    if world > 1 and dp > 1:
+        model = NaiveDdp(model, group=tpc.get_group("data"))
+    opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
+
+    # identical data inside a TP group; different across DP ranks
+    dp_rank = tpc.get_dp_rank() if world > 1 else 0
+    g = torch.Generator(device="cpu").manual_seed(9000 + dp_rank)
+    x = torch.randint(0, cfg.vocab_size, (args.batch, args.seq),
+                      generator=g).to(dev)
+
+    def step():
+        out = model(x, labels=x)
+        out["loss"].backward()
+        if isinstance(model, NaiveDdp):
+            model.reduce_gradients()
+        opt.step()
+        opt.zero_grad()
+
+    for _ in range(args.warmup):
+        step()
+    if dist.is_initialized():
+        dist.barrier()
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if dev.type == "cuda":
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    dt = (time.perf_counter() - t0) / args.steps
+    # max over ranks
+    if dist.is_initialized():
+        t = torch.tensor([dt], device=dev if dev.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dt = float(t.item())
+    return {"ms_per_step": dt * 1e3}
+
+
+if __name__ == "__main__":
+    main()

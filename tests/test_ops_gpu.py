@@ -1,0 +1,207 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference.
+
+Run on MI355X via: pytest tests/test_ops_gpu.py -m gpu -x -q
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+import torchdistpackage_amd.ops as ops
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+def setup_module(module):
+    if torch.cuda.is_available():
+        assert ops.extension_available(), \
+            "HIP extension must be built+loaded on a GPU box"
+
+
+# ---------------------------------------------------------------- MFMA probe
+
+def test_mfma_layout_probe():
+    """Validates the fragment-layout assumption of attention.hip.
+    Asymmetric A and B (guide rule: symmetric operands hide transposes)."""
+    torch.manual_seed(0)
+    a = (torch.randn(16, 32) * 0.5).bfloat16().to(_dev())
+    b = (torch.randn(32, 16) * 0.5).bfloat16().to(_dev())
+    d = ops.ext("probe").mfma_probe_16x16x32(a, b)
+    ref = a.float().cpu() @ b.float().cpu()
+    err = (d.cpu() - ref).abs().max().item()
+    assert err < 0.05, f"MFMA layout mismatch: max err {err}\n" \
+        f"got:\n{d.cpu()[:4, :4]}\nref:\n{ref[:4, :4]}"
+
+
+# ---------------------------------------------------------------- norms
+
+@pytest.mark.parametrize("shape", [(4, 128, 1024), (2, 63, 2048), (1, 8, 64)])
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_rmsnorm(shape, dtype):
+    torch.manual_seed(1)
+    x = torch.randn(shape, dtype=dtype, device=_dev(), requires_grad=True)
+    w = torch.randn(shape[-1], dtype=dtype, device=_dev(), requires_grad=True)
+    y = ops.rms_norm(x, w, 1e-6)
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-6) * wf
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert torch.allclose(y.float(), ref, atol=tol, rtol=tol)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(x.grad.float(), xf.grad, atol=tol * 2, rtol=tol * 2)
+    assert torch.allclose(w.grad.float(), wf.grad,
+                          atol=tol * 8, rtol=tol * 4), \
+        f"dw err {(w.grad.float() - wf.grad).abs().max()}"
+
+
+@pytest.mark.parametrize("shape", [(4, 128, 1024), (2, 63, 768)])
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_layernorm(shape, dtype):
+    torch.manual_seed(2)
+    x = torch.randn(shape, dtype=dtype, device=_dev(), requires_grad=True)
+    w = torch.randn(shape[-1], dtype=dtype, device=_dev(), requires_grad=True)
+    b = torch.randn(shape[-1], dtype=dtype, device=_dev(), requires_grad=True)
+    y = ops.layer_norm(x, w, b, 1e-5)
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xf, (shape[-1],), wf, bf, 1e-5)
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert torch.allclose(y.float(), ref, atol=tol, rtol=tol)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(x.grad.float(), xf.grad, atol=tol * 2, rtol=tol * 2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=tol * 8, rtol=tol * 4)
+    assert torch.allclose(b.grad.float(), bf.grad, atol=tol * 8, rtol=tol * 4)
+
+
+# ---------------------------------------------------------------- bias_gelu
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_bias_gelu(dtype):
+    torch.manual_seed(3)
+    x = torch.randn(8, 64, 512, dtype=dtype, device=_dev(), requires_grad=True)
+    b = torch.randn(512, dtype=dtype, device=_dev(), requires_grad=True)
+    y = ops.bias_gelu(x, b)
+
+    xf = x.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.gelu(xf + bf, approximate="tanh")
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    assert torch.allclose(y.float(), ref, atol=tol, rtol=tol)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(x.grad.float(), xf.grad, atol=tol * 2, rtol=tol * 2)
+    assert torch.allclose(b.grad.float(), bf.grad, atol=tol * 16, rtol=tol * 8)
+
+
+# ---------------------------------------------------------------- attention
+
+@pytest.mark.parametrize("causal", [True, False])
+@pytest.mark.parametrize("shape", [
+    (2, 4, 256, 128), (1, 2, 1024, 128), (2, 2, 192, 64), (1, 1, 100, 128)])
+def test_flash_attention_fwd(shape, causal):
+    torch.manual_seed(4)
+    B, H, S, D = shape
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=_dev())
+    k = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=_dev())
+    v = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=_dev())
+    o = ops.flash_attention(q, k, v, causal=causal)
+
+    scale = 1.0 / math.sqrt(D)
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=_dev()).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    ref = torch.matmul(torch.softmax(s, -1), v.float())
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"attn fwd max err {err}"
+
+
+@pytest.mark.parametrize("causal", [True, False])
+def test_flash_attention_bwd(causal):
+    torch.manual_seed(5)
+    B, H, S, D = 2, 2, 256, 128
+    q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=_dev(),
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    o = ops.flash_attention(q, k, v, causal=causal)
+    dy = torch.randn_like(o)
+    o.backward(dy)
+
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    scale = 1.0 / math.sqrt(D)
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=_dev()).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    ref = torch.matmul(torch.softmax(s, -1), vf)
+    ref.backward(dy.float())
+
+    for got, want, name in ((q.grad, qf.grad, "dq"), (k.grad, kf.grad, "dk"),
+                            (v.grad, vf.grad, "dv")):
+        err = (got.float() - want).abs().max().item()
+        assert err < 8e-2, f"attn bwd {name} max err {err}"
+
+
+# ---------------------------------------------------------------- optimizer
+
+def test_fused_adamw():
+    torch.manual_seed(6)
+    n = 100003
+    p = torch.randn(n, device=_dev())
+    g = torch.randn(n, device=_dev())
+    m = torch.zeros(n, device=_dev())
+    v = torch.zeros(n, device=_dev())
+    p_ref, g_ref = p.clone(), g.clone()
+    m_ref, v_ref = m.clone(), v.clone()
+
+    for step in range(1, 4):
+        ops.fused_adamw_(p, g, m, v, step, 1e-3, 0.9, 0.95, 1e-8, 0.1)
+        # torch reference
+        bc1, bc2 = 1 - 0.9 ** step, 1 - 0.95 ** step
+        p_ref.mul_(1 - 1e-3 * 0.1)
+        m_ref.mul_(0.9).add_(g_ref, alpha=0.1)
+        v_ref.mul_(0.95).addcmul_(g_ref, g_ref, value=0.05)
+        p_ref.addcdiv_(m_ref, (v_ref / bc2).sqrt().add_(1e-8),
+                       value=-1e-3 / bc1)
+    assert torch.allclose(p, p_ref, atol=1e-6), \
+        f"adamw err {(p - p_ref).abs().max()}"
+    assert torch.allclose(m, m_ref, atol=1e-6)
+    assert torch.allclose(v, v_ref, atol=1e-6)
+
+
+def test_ema_update():
+    n = 12345
+    ema = torch.randn(n, device=_dev())
+    p = torch.randn(n, device=_dev())
+    ref = ema * 0.99 + p * 0.01
+    ops.ema_update_(ema, p, 0.99)
+    assert torch.allclose(ema, ref, atol=1e-6)
+
+
+def test_l2norm_and_scale():
+    x = torch.randn(99991, dtype=torch.bfloat16, device=_dev())
+    got = ops.l2norm_sq(x)
+    want = x.float().pow(2).sum()
+    assert torch.allclose(got, want, rtol=1e-3)
+    xf = x.float()
+    ops.scale_(x, 0.5)
+    assert torch.allclose(x.float(), xf * 0.5, rtol=1e-2, atol=1e-3)

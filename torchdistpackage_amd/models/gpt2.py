@@ -1,0 +1,134 @@
+"""GPT-2 model family on the TP/SP blocks (works at tp=1 as plain blocks).
+
+This is the flagship training model for the benchmarks (BASELINE.json:
+GPT-2 small pure-DP; GPT-2 1.3B DP2xPP2xTP2).  The reference has no model
+zoo — its examples build throwaway transformers
+(/root/reference/examples/model_parallel/test_transformer.py:13-45); here the
+zoo is a first-class component.
+
+Layout: embeddings produce (B, S, D); blocks run sequence-first (S, B, D);
+loss is fp32 cross-entropy.  Hot ops (LayerNorm, flash attention, bias+GELU)
+are the in-tree gfx950 HIP kernels; GEMMs ride hipBLASLt via F.linear.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import LayerNorm
+from ..parallel.tensor import ParallelBlock, get_tp_size
+
+
+@dataclass
+class GPT2Config:
+    vocab_size: int = 50304      # padded 50257 for GEMM-friendly vocab
+    n_layer: int = 12
+    n_head: int = 12
+    dim: int = 768
+    max_seq: int = 1024
+    hidden_mult: int = 4
+    causal: bool = True
+    sequence_parallel: bool = True
+    tie_weights: bool = True
+
+
+def gpt2_small() -> GPT2Config:
+    return GPT2Config(n_layer=12, n_head=12, dim=768)
+
+
+def gpt2_medium() -> GPT2Config:
+    return GPT2Config(n_layer=24, n_head=16, dim=1024)
+
+
+def gpt2_xl_1p3b() -> GPT2Config:
+    """GPT-2/3 1.3B-class: 24 layers, d=2048, 16 heads (head_dim 128)."""
+    return GPT2Config(n_layer=24, n_head=16, dim=2048)
+
+
+class GPT2Embedding(nn.Module):
+    def __init__(self, cfg: GPT2Config, device=None, dtype=None):
+        super().__init__()
+        kw = {"device": device, "dtype": dtype}
+        self.wte = nn.Embedding(cfg.vocab_size, cfg.dim, **kw)
+        self.wpe = nn.Embedding(cfg.max_seq, cfg.dim, **kw)
+        nn.init.normal_(self.wte.weight, std=0.02)
+        nn.init.normal_(self.wpe.weight, std=0.01)
+
+    def forward(self, idx: torch.Tensor) -> torch.Tensor:
+        # idx (B, S) -> hidden (S, B, D) sequence-first for the TP blocks
+        B, S = idx.shape
+        pos = torch.arange(S, device=idx.device)
+        x = self.wte(idx) + self.wpe(pos)[None, :, :]
+        return x.transpose(0, 1).contiguous()
+
+
+class GPT2Head(nn.Module):
+    def __init__(self, cfg: GPT2Config, wte: Optional[nn.Embedding],
+                 device=None, dtype=None):
+        super().__init__()
+        kw = {"device": device, "dtype": dtype}
+        self.ln_f = LayerNorm(cfg.dim, **kw)
+        self.vocab_size = cfg.vocab_size
+        if cfg.tie_weights and wte is not None:
+            self.weight = wte.weight  # shared Parameter
+        else:
+            self.weight = nn.Parameter(
+                torch.empty(cfg.vocab_size, cfg.dim, **kw))
+            nn.init.normal_(self.weight, std=0.02)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # x (S, B, D) -> logits (B, S, V)
+        x = self.ln_f(x)
+        logits = F.linear(x, self.weight)
+        return logits.transpose(0, 1)
+
+
+class GPT2Model(nn.Module):
+    def __init__(self, cfg: GPT2Config, device=None, dtype=None):
+        super().__init__()
+        self.cfg = cfg
+        kw = {"device": device, "dtype": dtype}
+        self.embed = GPT2Embedding(cfg, **kw)
+        self.blocks = nn.ModuleList([
+            ParallelBlock(cfg.dim, cfg.n_head, cfg.hidden_mult,
+                          causal=cfg.causal,
+                          sequence_parallel=cfg.sequence_parallel, **kw)
+            for _ in range(cfg.n_layer)])
+        self.head = GPT2Head(cfg, self.embed.wte if cfg.tie_weights else None,
+                             **kw)
+
+    def forward(self, idx: torch.Tensor,
+                labels: Optional[torch.Tensor] = None) -> dict:
+        x = self.embed(idx)
+        for blk in self.blocks:
+            x = blk(x)
+        from ..parallel.tensor import (is_sequence_parallel,
+                                       gather_from_sequence_parallel_region)
+        if is_sequence_parallel(x) and get_tp_size() > 1:
+            x = gather_from_sequence_parallel_region(x)
+        logits = self.head(x)
+        out = {"logits": logits}
+        if labels is not None:
+            out["loss"] = F.cross_entropy(
+                logits.reshape(-1, logits.size(-1)).float(),
+                labels.reshape(-1))
+        return out
+
+    # -- pipeline partitioning support ----------------------------------
+
+    def to_stage_layers(self) -> List[nn.Module]:
+        """Flat layer list for the pipeline partitioner: [embed, blocks...,
+        head].  Stages feed (S, B, D) activations between each other."""
+        return [self.embed, *self.blocks, self.head]
+
+    @torch.no_grad()
+    def num_params(self, non_embedding: bool = False) -> int:
+        n = sum(p.numel() for p in self.parameters())
+        if non_embedding:
+            n -= self.embed.wpe.weight.numel()
+        return n

@@ -1,0 +1,345 @@
+"""Hand-written gfx950 HIP ops for the transformer hot path.
+
+These replace the implicit torch-op hot path the reference relies on
+(layernorm, naive SDPA, GELU, Adam step, EMA update,
+grad-norm/clip) with CDNA4 kernels (MFMA for attention, LDS-tiled /
+vectorized bf16 for the memory-bound ops).
+
+Dispatch policy:
+- CUDA (= HIP/MI355X) tensors: the in-tree extension ``_tdpa_hip`` MUST be
+  loaded — a missing extension raises instead of silently falling back to
+  eager torch (the silent-fallback trap the build rules call out).
+- CPU tensors: plain-torch reference implementations (used by the gloo CI
+  tests, and as the numerics oracle for the GPU kernels).
+"""
+
+from __future__ import annotations
+
+import math
+import os
+from typing import List, Optional
+
+import torch
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    import glob
+    import importlib.util
+    here = os.path.dirname(os.path.abspath(__file__))
+    cands = glob.glob(os.path.join(here, "_tdpa_hip*.so"))
+    if not cands:
+        _EXT_ERR = (f"in-tree HIP extension not found under {here}; "
+                    "run `python -m torchdistpackage_amd.ops.build` "
+                    "(or __graft_entry__.build())")
+        return None
+    spec = importlib.util.spec_from_file_location("_tdpa_hip", cands[0])
+    mod = importlib.util.module_from_spec(spec)
+    try:
+        spec.loader.exec_module(mod)
+    except Exception as e:  # loud, not silent
+        _EXT_ERR = f"failed to load {cands[0]}: {e}"
+        return None
+    _EXT = mod
+    return _EXT
+
+
+def ext(required_for: str = "op"):
+    """Return the extension module; raise if a GPU op needs it and it's
+    missing (no silent eager fallback on GPU)."""
+    m = _load_extension()
+    if m is None:
+        raise RuntimeError(
+            f"HIP extension required for {required_for} on GPU but "
+            f"unavailable: {_EXT_ERR}")
+    return m
+
+
+def extension_available() -> bool:
+    return _load_extension() is not None
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        if x.is_cuda:
+            y, rstd = ext("rms_norm").rmsnorm_fwd(x.contiguous(), weight, eps)
+        else:
+            xf = x.float()
+            rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+            y = (xf * rstd * weight.float()).to(x.dtype)
+            rstd = rstd.squeeze(-1)
+        ctx.save_for_backward(x, weight, rstd)
+        ctx.eps = eps
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd = ctx.saved_tensors
+        if x.is_cuda:
+            dx, dw = ext("rms_norm").rmsnorm_bwd(
+                dy.contiguous(), x.contiguous(), weight, rstd)
+        else:
+            xf, dyf, wf = x.float(), dy.float(), weight.float()
+            r = rstd.unsqueeze(-1)
+            xhat = xf * r
+            wdy = dyf * wf
+            # dx = r * (wdy - xhat * mean(wdy * xhat))
+            m = (wdy * xhat).mean(-1, keepdim=True)
+            dx = (r * (wdy - xhat * m)).to(x.dtype)
+            dw = (dyf * xhat).reshape(-1, x.shape[-1]).sum(0).to(weight.dtype)
+        return dx, dw, None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6):
+    return _RMSNormFn.apply(x, weight, eps)
+
+
+class RMSNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-6, device=None, dtype=None):
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(
+            torch.ones(dim, device=device, dtype=dtype))
+
+    def forward(self, x):
+        return rms_norm(x, self.weight, self.eps)
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm
+# ---------------------------------------------------------------------------
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        if x.is_cuda:
+            y, mean, rstd = ext("layer_norm").layernorm_fwd(
+                x.contiguous(), weight, bias, eps)
+        else:
+            xf = x.float()
+            mean = xf.mean(-1, keepdim=True)
+            var = xf.var(-1, unbiased=False, keepdim=True)
+            rstd = torch.rsqrt(var + eps)
+            y = ((xf - mean) * rstd * weight.float() + bias.float()).to(x.dtype)
+            mean, rstd = mean.squeeze(-1), rstd.squeeze(-1)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        if x.is_cuda:
+            dx, dw, db = ext("layer_norm").layernorm_bwd(
+                dy.contiguous(), x.contiguous(), weight, mean, rstd)
+        else:
+            xf, dyf, wf = x.float(), dy.float(), weight.float()
+            mu, r = mean.unsqueeze(-1), rstd.unsqueeze(-1)
+            xhat = (xf - mu) * r
+            wdy = dyf * wf
+            m1 = wdy.mean(-1, keepdim=True)
+            m2 = (wdy * xhat).mean(-1, keepdim=True)
+            dx = (r * (wdy - m1 - xhat * m2)).to(x.dtype)
+            dw = (dyf * xhat).reshape(-1, x.shape[-1]).sum(0).to(weight.dtype)
+            db = dyf.reshape(-1, x.shape[-1]).sum(0).to(weight.dtype)
+        return dx, dw, db, None
+
+
+def layer_norm(x, weight, bias, eps: float = 1e-5):
+    return _LayerNormFn.apply(x, weight, bias, eps)
+
+
+class LayerNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5, device=None, dtype=None):
+        super().__init__()
+        self.eps = eps
+        kw = {"device": device, "dtype": dtype}
+        self.weight = torch.nn.Parameter(torch.ones(dim, **kw))
+        self.bias = torch.nn.Parameter(torch.zeros(dim, **kw))
+
+    def forward(self, x):
+        return layer_norm(x, self.weight, self.bias, self.eps)
+
+
+# ---------------------------------------------------------------------------
+# fused bias + GELU (tanh approximation, GPT-2 convention)
+# ---------------------------------------------------------------------------
+
+_GELU_C = math.sqrt(2.0 / math.pi)
+
+
+def _gelu_tanh(u):
+    return 0.5 * u * (1.0 + torch.tanh(_GELU_C * (u + 0.044715 * u.pow(3))))
+
+
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        ctx.has_bias = bias is not None
+        if x.is_cuda:
+            y = ext("bias_gelu").bias_gelu_fwd(
+                x.contiguous(),
+                bias if bias is not None else torch.empty(
+                    0, dtype=x.dtype, device=x.device))
+            ctx.save_for_backward(x, bias if bias is not None else
+                                  torch.empty(0, dtype=x.dtype, device=x.device))
+        else:
+            u = x.float() + (bias.float() if bias is not None else 0.0)
+            y = _gelu_tanh(u).to(x.dtype)
+            ctx.save_for_backward(x, bias if bias is not None else
+                                  torch.empty(0, dtype=x.dtype))
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, bias = ctx.saved_tensors
+        if x.is_cuda:
+            dx = ext("bias_gelu").bias_gelu_bwd(dy.contiguous(),
+                                                x.contiguous(), bias)
+        else:
+            u = x.float() + (bias.float() if ctx.has_bias else 0.0)
+            t = torch.tanh(_GELU_C * (u + 0.044715 * u.pow(3)))
+            du = 0.5 * (1 + t) + 0.5 * u * (1 - t * t) * _GELU_C * \
+                (1 + 3 * 0.044715 * u.pow(2))
+            dx = (dy.float() * du).to(x.dtype)
+        dbias = None
+        if ctx.has_bias:
+            dbias = dx.reshape(-1, x.shape[-1]).sum(0).to(x.dtype)
+        return dx, dbias
+
+
+def bias_gelu(x, bias=None):
+    return _BiasGeluFn.apply(x, bias)
+
+
+# ---------------------------------------------------------------------------
+# flash attention (bf16, causal/full, head_dim 64/128)
+# ---------------------------------------------------------------------------
+
+class _FlashAttentionFn(torch.autograd.Function):
+    """q,k,v: (B, H, S, D) contiguous.  GPU path = in-tree HIP flash kernel
+    (blockwise online-softmax, spec: reference explore/flash-attn/
+    tile_attn.py:100-212); CPU path = exact math attention in fp32."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        if q.is_cuda:
+            o, lse = ext("flash_attention").attn_fwd(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                bool(causal), float(scale))
+        else:
+            qf, kf, vf = q.float(), k.float(), v.float()
+            s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+            if causal:
+                S = q.shape[-2]
+                mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
+                                  device=q.device).tril_()
+                s = s.masked_fill(~mask, float("-inf"))
+            lse = torch.logsumexp(s, dim=-1)
+            p = torch.softmax(s, dim=-1)
+            o = torch.matmul(p, vf).to(q.dtype)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        if q.is_cuda:
+            # Recompute-based backward from the saved LSE: P is rebuilt as
+            # exp(S - lse) without a second softmax pass; the five GEMMs run
+            # on hipBLASLt in bf16.  (Interim composite — the fused HIP bwd
+            # kernel replaces this; the fwd already runs the HIP kernel.)
+            ext("flash_attention")  # assert extension presence loudly
+            dof = do.contiguous()
+            s = torch.matmul(q, k.transpose(-1, -2)).float() * ctx.scale
+            if ctx.causal:
+                S = q.shape[-2]
+                mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
+                                  device=q.device).tril_()
+                s = s.masked_fill(~mask, float("-inf"))
+            p = torch.exp(s - lse.unsqueeze(-1)).to(q.dtype)
+            dv = torch.matmul(p.transpose(-1, -2), dof)
+            dp = torch.matmul(dof, v.transpose(-1, -2)).float()
+            d = (dof.float() * o.float()).sum(-1, keepdim=True)
+            ds = (p.float() * (dp - d) * ctx.scale).to(q.dtype)
+            dq = torch.matmul(ds, k)
+            dk = torch.matmul(ds.transpose(-1, -2), q)
+        else:
+            qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
+            s = torch.matmul(qf, kf.transpose(-1, -2)) * ctx.scale
+            if ctx.causal:
+                S = q.shape[-2]
+                mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
+                                  device=q.device).tril_()
+                s = s.masked_fill(~mask, float("-inf"))
+            p = torch.softmax(s, dim=-1)
+            dv = torch.matmul(p.transpose(-1, -2), dof)
+            dp = torch.matmul(dof, vf.transpose(-1, -2))
+            d = (dp * p).sum(-1, keepdim=True)
+            ds = p * (dp - d) * ctx.scale
+            dq = torch.matmul(ds, kf)
+            dk = torch.matmul(ds.transpose(-1, -2), qf)
+            dq, dk, dv = dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
+        return dq, dk, dv, None, None
+
+
+def flash_attention(q, k, v, causal: bool = True, scale: float = None):
+    return _FlashAttentionFn.apply(q, k, v, causal, scale)
+
+
+# ---------------------------------------------------------------------------
+# fused optimizer / EMA / grad utilities (flat-tensor kernels)
+# ---------------------------------------------------------------------------
+
+def fused_adamw_(param: torch.Tensor, grad: torch.Tensor,
+                 exp_avg: torch.Tensor, exp_avg_sq: torch.Tensor,
+                 step: int, lr: float, beta1: float, beta2: float,
+                 eps: float, weight_decay: float):
+    """AdamW update on (flat) fp32 tensors, one fused kernel pass on GPU."""
+    if param.is_cuda:
+        ext("fused_adamw").adamw_step(param, grad, exp_avg, exp_avg_sq,
+                                      step, lr, beta1, beta2, eps,
+                                      weight_decay)
+        return
+    bc1 = 1 - beta1 ** step
+    bc2 = 1 - beta2 ** step
+    param.mul_(1 - lr * weight_decay)
+    exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+    denom = (exp_avg_sq / bc2).sqrt_().add_(eps)
+    param.addcdiv_(exp_avg, denom, value=-lr / bc1)
+
+
+def ema_update_(ema: torch.Tensor, param: torch.Tensor, decay: float):
+    """ema = decay*ema + (1-decay)*param, fused."""
+    if ema.is_cuda:
+        ext("ema_update").ema_update(ema, param, decay)
+        return
+    ema.lerp_(param.to(ema.dtype), 1.0 - decay)
+
+
+def l2norm_sq(t: torch.Tensor) -> torch.Tensor:
+    """Sum of squares (fp32 accumulate) of one (flat) tensor."""
+    if t.is_cuda:
+        return ext("l2norm").l2norm_sq(t)
+    return t.float().pow(2).sum()
+
+
+def scale_(t: torch.Tensor, scale: float):
+    if t.is_cuda:
+        ext("scale").scale_inplace(t, scale)
+        return
+    t.mul_(scale)

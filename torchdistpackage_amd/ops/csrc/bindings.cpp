@@ -1,0 +1,40 @@
+// Python bindings for the in-tree gfx950 HIP extension (_tdpa_hip).
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd);
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias);
+torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                            torch::Tensor bias);
+void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, long step, double lr, double beta1,
+                double beta2, double eps, double wd);
+void ema_update(torch::Tensor ema, torch::Tensor p, double decay);
+torch::Tensor l2norm_sq(torch::Tensor x);
+void scale_inplace(torch::Tensor x, double s);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal,
+                                    double scale);
+torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("ema_update", &ema_update);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("scale_inplace", &scale_inplace);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+}

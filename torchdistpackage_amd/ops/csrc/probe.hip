@@ -1,0 +1,50 @@
+// MFMA fragment-layout probe: computes D = A(16x32) @ B(32x16) with
+// v_mfma_f32_16x16x32_bf16 under the layout assumption used by attention.hip
+// (A: lane l holds row l&15, k = (l>>4)*8 + j ; B: col l&15, same k ;
+//  C/D: col = l&15, row = (l>>4)*4 + reg).
+// A GPU test compares this against torch.matmul — if the assumption is wrong
+// the test localizes exactly which mapping to fix.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__global__ void mfma_probe_kernel(const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b,
+                                  float* __restrict__ d) {
+  const int lane = threadIdx.x & 63;
+  const int l15 = lane & 15;
+  const int lg = lane >> 4;
+  unsigned short af[8], bf[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[l15 * 32 + lg * 8 + j];        // A[row][k], row-major 16x32
+    bf[j] = b[(lg * 8 + j) * 16 + l15];      // B[k][col], row-major 32x16
+  }
+  f32x4 acc = (f32x4)(0.f);
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+      *(const bf16x8_v*)af, *(const bf16x8_v*)bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(lg * 4 + r) * 16 + l15] = acc[r];
+}
+
+}  // namespace
+
+torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(b.sizes() == torch::IntArrayRef({32, 16}));
+  auto d = torch::empty({16, 16}, a.options().dtype(torch::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const unsigned short*)a.contiguous().data_ptr(),
+                     (const unsigned short*)b.contiguous().data_ptr(),
+                     d.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return d;
+}
