@@ -258,25 +258,8 @@ class _FlashAttentionFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         if q.is_cuda:
-            # Recompute-based backward from the saved LSE: P is rebuilt as
-            # exp(S - lse) without a second softmax pass; the five GEMMs run
-            # on hipBLASLt in bf16.  (Interim composite — the fused HIP bwd
-            # kernel replaces this; the fwd already runs the HIP kernel.)
-            ext("flash_attention")  # assert extension presence loudly
-            dof = do.contiguous()
-            s = torch.matmul(q, k.transpose(-1, -2)).float() * ctx.scale
-            if ctx.causal:
-                S = q.shape[-2]
-                mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
-                                  device=q.device).tril_()
-                s = s.masked_fill(~mask, float("-inf"))
-            p = torch.exp(s - lse.unsqueeze(-1)).to(q.dtype)
-            dv = torch.matmul(p.transpose(-1, -2), dof)
-            dp = torch.matmul(dof, v.transpose(-1, -2)).float()
-            d = (dof.float() * o.float()).sum(-1, keepdim=True)
-            ds = (p.float() * (dp - d) * ctx.scale).to(q.dtype)
-            dq = torch.matmul(ds, k)
-            dk = torch.matmul(ds.transpose(-1, -2), q)
+            dq, dk, dv = ext("flash_attention").attn_bwd(
+                do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale)
         else:
             qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
             s = torch.matmul(qf, kf.transpose(-1, -2)) * ctx.scale

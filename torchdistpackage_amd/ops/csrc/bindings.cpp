@@ -22,6 +22,10 @@ void scale_inplace(torch::Tensor x, double s);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal,
                                     double scale);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -36,5 +40,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_inplace", &scale_inplace);
   m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
 }

@@ -198,6 +198,218 @@ int pick_grid(long rows) {
   return (int)(g > 0 ? g : 1);
 }
 
+// ------------------- vectorized bf16 variants (D % 8 == 0) ----------------
+// 16 B/lane loads (guideline 13: scalar bf16 is 2-2.5x slower).
+
+typedef ushort4 u4;
+struct f8 { float v[8]; };
+
+DEVINL f8 load8f(const unsigned short* p) {
+  u4 a = *(const u4*)p;
+  u4 b = *(const u4*)(p + 4);
+  f8 r;
+  r.v[0] = bf2f(a.x); r.v[1] = bf2f(a.y); r.v[2] = bf2f(a.z); r.v[3] = bf2f(a.w);
+  r.v[4] = bf2f(b.x); r.v[5] = bf2f(b.y); r.v[6] = bf2f(b.z); r.v[7] = bf2f(b.w);
+  return r;
+}
+
+DEVINL void store8f(unsigned short* p, const f8& r) {
+  u4 a, b;
+  a.x = f2bf(r.v[0]); a.y = f2bf(r.v[1]); a.z = f2bf(r.v[2]); a.w = f2bf(r.v[3]);
+  b.x = f2bf(r.v[4]); b.y = f2bf(r.v[5]); b.z = f2bf(r.v[6]); b.w = f2bf(r.v[7]);
+  *(u4*)p = a;
+  *(u4*)(p + 4) = b;
+}
+
+__global__ void rmsnorm_fwd_bf16v8(const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ w,
+                                   unsigned short* __restrict__ y,
+                                   float* __restrict__ rstd,
+                                   int rows, int D, float eps) {
+  __shared__ float lds[BLOCK / WAVE];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* yr = y + (long)row * D;
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 v = load8f(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ss += v.v[j] * v.v[j];
+    }
+    ss = block_sum<BLOCK>(ss, lds);
+    float r = rsqrtf(ss / D + eps);
+    if (threadIdx.x == 0) rstd[row] = r;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 v = load8f(xr + i);
+      f8 wv = load8f(w + i);
+      f8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out.v[j] = v.v[j] * r * wv.v[j];
+      store8f(yr + i, out);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void rmsnorm_bwd_bf16v8(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ w,
+                                   const float* __restrict__ rstd,
+                                   unsigned short* __restrict__ dx,
+                                   float* __restrict__ dw_partial,
+                                   int rows, int D) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* dw_lds = (float*)smem;
+  float* scratch = dw_lds + D;
+  for (int i = threadIdx.x; i < D; i += BLOCK) dw_lds[i] = 0.f;
+  __syncthreads();
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + (long)row * D;
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* dxr = dx + (long)row * D;
+    float r = rstd[row];
+    float dot = 0.f;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 xv = load8f(xr + i), dyv = load8f(dyr + i), wv = load8f(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dot += wv.v[j] * dyv.v[j] * xv.v[j] * r;
+    }
+    dot = block_sum<BLOCK>(dot, scratch) / D;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 xv = load8f(xr + i), dyv = load8f(dyr + i), wv = load8f(w + i);
+      f8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xhat = xv.v[j] * r;
+        out.v[j] = r * (wv.v[j] * dyv.v[j] - xhat * dot);
+        dw_lds[i + j] += dyv.v[j] * xhat;
+      }
+      store8f(dxr + i, out);
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < D; i += BLOCK)
+    atomicAdd(&dw_partial[i], dw_lds[i]);
+}
+
+__global__ void layernorm_fwd_bf16v8(const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     const unsigned short* __restrict__ b,
+                                     unsigned short* __restrict__ y,
+                                     float* __restrict__ mean,
+                                     float* __restrict__ rstd,
+                                     int rows, int D, float eps) {
+  __shared__ float lds[2 * BLOCK / WAVE];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* yr = y + (long)row * D;
+    float s = 0.f, ss = 0.f;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 v = load8f(xr + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { s += v.v[j]; ss += v.v[j] * v.v[j]; }
+    }
+    // two block sums sharing one pass: use separate scratch halves
+    {
+      const int wid = threadIdx.x / WAVE;
+      float sw = wave_sum(s), ssw = wave_sum(ss);
+      if ((threadIdx.x & (WAVE - 1)) == 0) {
+        lds[wid] = sw;
+        lds[BLOCK / WAVE + wid] = ssw;
+      }
+      __syncthreads();
+      s = 0.f; ss = 0.f;
+#pragma unroll
+      for (int i = 0; i < BLOCK / WAVE; ++i) {
+        s += lds[i];
+        ss += lds[BLOCK / WAVE + i];
+      }
+    }
+    float mu = s / D;
+    float var = ss / D - mu * mu;
+    float r = rsqrtf(var + eps);
+    if (threadIdx.x == 0) { mean[row] = mu; rstd[row] = r; }
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 v = load8f(xr + i), wv = load8f(w + i), bv = load8f(b + i);
+      f8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        out.v[j] = (v.v[j] - mu) * r * wv.v[j] + bv.v[j];
+      store8f(yr + i, out);
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void layernorm_bwd_bf16v8(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     unsigned short* __restrict__ dx,
+                                     float* __restrict__ dw_partial,
+                                     float* __restrict__ db_partial,
+                                     int rows, int D) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* dw_lds = (float*)smem;
+  float* db_lds = dw_lds + D;
+  float* scratch = db_lds + D;   // 2*BLOCK/WAVE floats
+  for (int i = threadIdx.x; i < D; i += BLOCK) { dw_lds[i] = 0.f; db_lds[i] = 0.f; }
+  __syncthreads();
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + (long)row * D;
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* dxr = dx + (long)row * D;
+    float mu = mean[row], r = rstd[row];
+    float m1 = 0.f, m2 = 0.f;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 xv = load8f(xr + i), dyv = load8f(dyr + i), wv = load8f(w + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xhat = (xv.v[j] - mu) * r;
+        float wdy = wv.v[j] * dyv.v[j];
+        m1 += wdy;
+        m2 += wdy * xhat;
+      }
+    }
+    {
+      const int wid = threadIdx.x / WAVE;
+      float a = wave_sum(m1), c = wave_sum(m2);
+      if ((threadIdx.x & (WAVE - 1)) == 0) {
+        scratch[wid] = a;
+        scratch[BLOCK / WAVE + wid] = c;
+      }
+      __syncthreads();
+      m1 = 0.f; m2 = 0.f;
+#pragma unroll
+      for (int i = 0; i < BLOCK / WAVE; ++i) {
+        m1 += scratch[i];
+        m2 += scratch[BLOCK / WAVE + i];
+      }
+    }
+    m1 /= D;
+    m2 /= D;
+    for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+      f8 xv = load8f(xr + i), dyv = load8f(dyr + i), wv = load8f(w + i);
+      f8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xhat = (xv.v[j] - mu) * r;
+        out.v[j] = r * (wv.v[j] * dyv.v[j] - m1 - xhat * m2);
+        dw_lds[i + j] += dyv.v[j] * xhat;
+        db_lds[i + j] += dyv.v[j];
+      }
+      store8f(dxr + i, out);
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < D; i += BLOCK) {
+    atomicAdd(&dw_partial[i], dw_lds[i]);
+    atomicAdd(&db_partial[i], db_lds[i]);
+  }
+}
+
 }  // namespace
 
 // ------------------------------------------------------------------ C++ API
@@ -212,7 +424,13 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
-  if (x.scalar_type() == torch::kBFloat16) {
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+    hipLaunchKernelGGL(rmsnorm_fwd_bf16v8, grid, block, 0, stream,
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (unsigned short*)y.data_ptr(), rstd.data_ptr<float>(),
+                       (int)rows, D, (float)eps);
+  } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(rmsnorm_fwd_kernel<bf16_t>, grid, block, 0, stream,
                        (const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
                        (bf16_t*)y.data_ptr(), rstd.data_ptr<float>(),
@@ -237,7 +455,14 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
   size_t lds = (D + BLOCK / WAVE) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16) {
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+    hipLaunchKernelGGL(rmsnorm_bwd_bf16v8, grid, block, lds, stream,
+                       (const unsigned short*)dy.data_ptr(),
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       rstd.data_ptr<float>(), (unsigned short*)dx.data_ptr(),
+                       dw_partial.data_ptr<float>(), (int)rows, D);
+  } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(rmsnorm_bwd_kernel<bf16_t>, grid, block, lds, stream,
                        (const bf16_t*)dy.data_ptr(), (const bf16_t*)x.data_ptr(),
                        (const bf16_t*)w.data_ptr(), rstd.data_ptr<float>(),
@@ -265,7 +490,15 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
   auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
-  if (x.scalar_type() == torch::kBFloat16) {
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+    hipLaunchKernelGGL(layernorm_fwd_bf16v8, grid, block, 0, stream,
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (const unsigned short*)b.data_ptr(),
+                       (unsigned short*)y.data_ptr(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       (int)rows, D, (float)eps);
+  } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(layernorm_fwd_kernel<bf16_t>, grid, block, 0, stream,
                        (const bf16_t*)x.data_ptr(), (const bf16_t*)w.data_ptr(),
                        (const bf16_t*)b.data_ptr(), (bf16_t*)y.data_ptr(),
@@ -293,8 +526,17 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto db_partial = torch::zeros({D}, x.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
-  size_t lds = (2 * D + BLOCK / WAVE) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16) {
+  size_t lds = (2 * D + 2 * (BLOCK / WAVE)) * sizeof(float);
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+    hipLaunchKernelGGL(layernorm_bwd_bf16v8, grid, block, lds, stream,
+                       (const unsigned short*)dy.data_ptr(),
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(),
+                       dw_partial.data_ptr<float>(),
+                       db_partial.data_ptr<float>(), (int)rows, D);
+  } else if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(layernorm_bwd_kernel<bf16_t>, grid, block, lds, stream,
                        (const bf16_t*)dy.data_ptr(), (const bf16_t*)x.data_ptr(),
                        (const bf16_t*)w.data_ptr(), mean.data_ptr<float>(),

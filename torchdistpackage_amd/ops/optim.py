@@ -1,19 +1,48 @@
-"""FusedAdamW: AdamW on the in-tree HIP kernel, with fp32 master weights for
-bf16 params.
+"""FusedAdamW: flat multi-tensor AdamW on the in-tree HIP kernel.
 
-Replaces the reference's reliance on stock torch.optim.Adam/AdamW
-(/root/reference/torchdistpackage/ddp/zero_optim.py:265, examples).  Per-param
-state lives in fp32; on GPU the update is one fused kernel per param tensor
-(param/exp_avg/exp_avg_sq read+write in a single HBM pass).
+Replaces the reference's stock torch.optim.Adam/AdamW
+(/root/reference/torchdistpackage/ddp/zero_optim.py:265, examples) with an
+MI355X-shaped update: per param-group, ALL state lives in single contiguous
+fp32 flats (master weights, exp_avg, exp_avg_sq, grad buffer), so a step is
+
+    1. one fused foreach-copy of grads into the fp32 grad flat (casts bf16)
+    2. ONE adamw kernel pass over the whole flat (4 reads + 3 writes of HBM)
+    3. one fused foreach-copy of updated master views back into the params
+
+instead of ~6 kernels x #params.  Profiling of the v0 per-param variant showed
+~1900 cast + ~1700 copy kernels per 4 steps dominating step time.
 """
 
 from __future__ import annotations
 
-from typing import Optional
+from typing import List
 
 import torch
 
 from . import fused_adamw_
+
+
+class _FlatGroup:
+    def __init__(self, params: List[torch.Tensor]):
+        self.params = params
+        n = sum(p.numel() for p in params)
+        dev = params[0].device
+        self.master = torch.empty(n, dtype=torch.float32, device=dev)
+        self.grad32 = torch.empty(n, dtype=torch.float32, device=dev)
+        self.exp_avg = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.exp_avg_sq = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.master_views = []
+        self.grad_views = []
+        off = 0
+        for p in params:
+            k = p.numel()
+            mv = self.master.narrow(0, off, k).view_as(p)
+            mv.copy_(p.detach().to(torch.float32))
+            self.master_views.append(mv)
+            self.grad_views.append(self.grad32.narrow(0, off, k).view_as(p))
+            off += k
+        self.all_fp32_inplace = all(
+            p.dtype == torch.float32 for p in params)
 
 
 class FusedAdamW(torch.optim.Optimizer):
@@ -22,6 +51,15 @@ class FusedAdamW(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self._flats: List[_FlatGroup] = []
+        self._step = 0
+        self._built = False
+
+    def _build(self):
+        for group in self.param_groups:
+            ps = [p for p in group["params"] if p.requires_grad]
+            group["_flat"] = _FlatGroup(ps) if ps else None
+        self._built = True
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -29,32 +67,53 @@ class FusedAdamW(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        if not self._built:
+            self._build()
+        self._step += 1
         for group in self.param_groups:
+            fg: _FlatGroup = group.get("_flat")
+            if fg is None:
+                continue
             lr = group["lr"]
             beta1, beta2 = group["betas"]
-            eps = group["eps"]
-            wd = group["weight_decay"]
-            for p in group["params"]:
-                if p.grad is None:
-                    continue
-                state = self.state[p]
-                if len(state) == 0:
-                    state["step"] = 0
-                    state["exp_avg"] = torch.zeros_like(
-                        p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(
-                        p, dtype=torch.float32)
-                    if p.dtype != torch.float32:
-                        state["master"] = p.detach().float().clone()
-                state["step"] += 1
-                grad = p.grad
-                if grad.dtype != torch.float32:
-                    grad = grad.float()
-                target = state.get("master", p)
-                fused_adamw_(target.view(-1), grad.contiguous().view(-1),
-                             state["exp_avg"].view(-1),
-                             state["exp_avg_sq"].view(-1),
-                             state["step"], lr, beta1, beta2, eps, wd)
-                if "master" in state:
-                    p.copy_(target.to(p.dtype))
+            grads = [p.grad if p.grad is not None
+                     else torch.zeros_like(p) for p in fg.params]
+            torch._foreach_copy_(fg.grad_views, grads)
+            fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
+                         self._step, lr, beta1, beta2, group["eps"],
+                         group["weight_decay"])
+            torch._foreach_copy_(fg.params, fg.master_views)
         return loss
+
+    def zero_grad(self, set_to_none: bool = True):
+        for group in self.param_groups:
+            for p in group["params"]:
+                if set_to_none:
+                    p.grad = None
+                elif p.grad is not None:
+                    p.grad.zero_()
+
+    def state_dict(self):
+        if not self._built:
+            self._build()
+        return {
+            "step": self._step,
+            "groups": [
+                {"master": g["_flat"].master if g["_flat"] else None,
+                 "exp_avg": g["_flat"].exp_avg if g["_flat"] else None,
+                 "exp_avg_sq": g["_flat"].exp_avg_sq if g["_flat"] else None}
+                for g in self.param_groups],
+        }
+
+    def load_state_dict(self, sd):
+        if not self._built:
+            self._build()
+        self._step = sd["step"]
+        for g, gsd in zip(self.param_groups, sd["groups"]):
+            fg = g.get("_flat")
+            if fg is None or gsd["master"] is None:
+                continue
+            fg.master.copy_(gsd["master"])
+            fg.exp_avg.copy_(gsd["exp_avg"])
+            fg.exp_avg_sq.copy_(gsd["exp_avg_sq"])
+            torch._foreach_copy_(fg.params, fg.master_views)
