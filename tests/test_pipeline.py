@@ -1,0 +1,179 @@
+"""1F1B pipeline tests (gloo, CPU): numeric oracle vs single-process run.
+
+Goes beyond the reference's smoke-only pipeline test
+(examples/model_parallel/test_pipeline.py — "it runs"): grads after a full
+1F1B iteration are compared against the same model run unpartitioned.
+world_size=3 covers the mid-stage path the reference's pp=2 testing missed
+(SURVEY.md known-bugs: pipeline_sched.py:129).
+"""
+
+import copy
+
+import pytest
+import torch
+import torch.nn as nn
+
+from tests.dist_helpers import run_distributed
+from torchdistpackage_amd.parallel.pipeline.partition import (
+    partition_balanced, partition_uniform, flatten_sequence)
+
+
+def test_partition_uniform():
+    assert partition_uniform(8, 2) == [[0, 4], [4, 8]]
+    assert partition_uniform(7, 2) == [[0, 4], [4, 7]]
+    assert partition_uniform(5, 4) == [[0, 2], [2, 3], [3, 4], [4, 5]]
+
+
+def test_partition_balanced():
+    layers = [nn.Linear(10, 10), nn.Linear(100, 100), nn.Linear(10, 10),
+              nn.Linear(10, 10)]
+    parts = partition_balanced(layers, 2)
+    # the 100x100 layer dominates; it should sit alone-ish
+    assert parts[0][1] == 2 and parts[1] == [2, 4]
+
+
+def test_flatten_sequence():
+    m = nn.Sequential(nn.Linear(2, 2),
+                      nn.Sequential(nn.ReLU(), nn.Linear(2, 2)))
+    flat = flatten_sequence(m)
+    assert len(flat) == 3
+
+
+def _make_layers(seed=0, depth=6, dim=32):
+    torch.manual_seed(seed)
+    return nn.Sequential(*[nn.Sequential(nn.Linear(dim, dim), nn.Tanh())
+                           for _ in range(depth)])
+
+
+def _pp_iteration(rank, world_size, num_microbatches=4):
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.pipeline import (forward_backward,
+                                                        partition_uniform)
+
+    tpc.setup_process_groups([("pipe", world_size)])
+    depth, dim, B = 6, 32, 8
+    full = _make_layers(seed=42, depth=depth, dim=dim)
+    ref = copy.deepcopy(full)
+
+    parts = partition_uniform(depth, world_size)
+    s, e = parts[rank]
+    stage = nn.Sequential(*list(full)[s:e])
+
+    torch.manual_seed(7)
+    x = torch.randn(B, dim)
+
+    def fwd_fn(stage_in):
+        out = stage(stage_in)
+        if tpc.is_last_in_pipeline_group():
+            # per-micro-batch loss normalized by num_microbatches
+            return out.pow(2).mean() / num_microbatches
+        return out
+
+    losses = forward_backward(fwd_fn, inputs=x,
+                              num_microbatches=num_microbatches,
+                              return_losses=True)
+
+    # oracle: same model, plain full-batch (mean over micro-batch losses ==
+    # full-batch loss since micro-batches are equal-size)
+    ref(x).pow(2).mean().backward()
+    ref_stage = nn.Sequential(*list(ref)[s:e])
+    for (n, p), (rn, rp) in zip(stage.named_parameters(),
+                                ref_stage.named_parameters()):
+        assert p.grad is not None, f"stage {rank} param {n} got no grad"
+        assert torch.allclose(p.grad, rp.grad, atol=1e-5), \
+            f"stage {rank} grad mismatch {n}: " \
+            f"{(p.grad - rp.grad).abs().max().item()}"
+    if tpc.is_last_in_pipeline_group():
+        total = sum(float(l) for l in losses)
+        ref_loss = float(ref(x).pow(2).mean())
+        assert abs(total - ref_loss) < 1e-5
+    return True
+
+
+def test_1f1b_pp2():
+    run_distributed(_pp_iteration, world_size=2)
+
+
+def test_1f1b_pp3():
+    run_distributed(_pp_iteration, world_size=3)
+
+
+def test_1f1b_pp2_many_microbatches():
+    run_distributed(_pp_iteration, world_size=2,
+                    kwargs={"num_microbatches": 8})
+
+
+def _pp_eval(rank, world_size):
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.pipeline import (forward_eval,
+                                                        partition_uniform)
+
+    tpc.setup_process_groups([("pipe", world_size)])
+    depth, dim, B = 4, 16, 4
+    full = _make_layers(seed=3, depth=depth, dim=dim)
+    parts = partition_uniform(depth, world_size)
+    s, e = parts[rank]
+    stage = nn.Sequential(*list(full)[s:e])
+    torch.manual_seed(5)
+    x = torch.randn(B, dim)
+    outs = forward_eval(lambda t: stage(t), inputs=x, num_microbatches=2)
+    if tpc.is_last_in_pipeline_group():
+        with torch.no_grad():
+            ref = full(x)
+        got = torch.cat(outs, dim=0)
+        assert torch.allclose(got, ref, atol=1e-6)
+    return True
+
+
+def test_forward_eval_pp2():
+    run_distributed(_pp_eval, world_size=2)
+
+
+def _pp_with_ddp(rank, world_size):
+    """dp2 x pp2 on 4 ranks: NaiveDdp with num_grad_acc_iter=num_microbatches
+    reduces only at the last micro-batch (reference Readme.md:56 claim)."""
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.ddp import NaiveDdp
+    from torchdistpackage_amd.parallel.pipeline import (forward_backward,
+                                                        partition_uniform)
+
+    tpc.setup_process_groups([("data", 2), ("pipe", 2)])
+    depth, dim, B, n_mb = 4, 16, 8, 4
+    full = _make_layers(seed=11, depth=depth, dim=dim)
+    ref = copy.deepcopy(full)
+    s, e = partition_uniform(depth, 2)[tpc.get_pp_rank()]
+    stage = nn.Sequential(*list(full)[s:e])
+    stage_ddp = NaiveDdp(stage, group=tpc.get_group("data"),
+                         num_grad_acc_iter=n_mb)
+
+    dp_rank = tpc.get_dp_rank()
+    torch.manual_seed(100 + dp_rank)
+    x = torch.randn(B, dim)
+
+    def fwd_fn(stage_in):
+        out = stage_ddp(stage_in)
+        if tpc.is_last_in_pipeline_group():
+            return out.pow(2).mean() / n_mb
+        return out
+
+    forward_backward(fwd_fn, inputs=x, num_microbatches=n_mb)
+    stage_ddp.reduce_gradients()
+
+    # oracle: average of both dp ranks' full-batch grads
+    for r in range(2):
+        torch.manual_seed(100 + r)
+        xr = torch.randn(B, dim)
+        ref(xr).pow(2).mean().backward()
+    ref_stage = nn.Sequential(*list(ref)[s:e])
+    for (n, p), (rn, rp) in zip(stage.named_parameters(),
+                                ref_stage.named_parameters()):
+        assert torch.allclose(p.grad, rp.grad / 2, atol=1e-5), \
+            f"{n}: {(p.grad - rp.grad / 2).abs().max().item()}"
+    return True
+
+
+def test_pp2_dp2_composition():
+    run_distributed(_pp_with_ddp, world_size=4)

@@ -1,0 +1,213 @@
+"""1F1B pipeline schedule.
+
+Capability parity with the reference scheduler
+(/root/reference/torchdistpackage/parallel/pipeline_parallel/pipeline_sched.py):
+non-linear-model-capable 1F1B with user-supplied per-stage ``fwd_fn`` (and
+optional ``bwd_fn``), micro-batch slicing of extra inputs, warmup =
+pp_size - pp_rank - 1 forwards, fused send/recv in the steady state, cooldown
+backwards, and ``forward_eval`` for inference pipelining.
+
+Fixes vs the reference: the ``is_first_in_pipeline_group`` missing-parens bug
+(pipeline_sched.py:129 — a truthy bound method that skips the warmup meta
+recv and breaks pp_size>=3) does not exist here; shape handshakes happen once
+per schedule (first micro-batch) and shapes are cached after.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional, Sequence
+
+import torch
+
+from ...dist.topo import tpc
+from . import p2p
+
+
+def _slice_microbatch(inputs, i: int, num_microbatches: int):
+    """Slice tensor / tuple-of-tensors along dim0 into micro-batch i."""
+    if inputs is None:
+        return None
+    if torch.is_tensor(inputs):
+        mb = inputs.shape[0] // num_microbatches
+        return inputs.narrow(0, i * mb, mb)
+    return type(inputs)(_slice_microbatch(t, i, num_microbatches)
+                        for t in inputs)
+
+
+def _default_bwd(output, grad_output, optimizer=None, scaler=None):
+    if grad_output is None:
+        if scaler is not None:
+            output = scaler.scale(output)
+        torch.autograd.backward(output)
+    else:
+        torch.autograd.backward(output, grad_tensors=grad_output)
+
+
+def forward_backward(fwd_fn: Callable,
+                     inputs=None,
+                     num_microbatches: int = 1,
+                     optimizer=None,
+                     bwd_fn: Optional[Callable] = None,
+                     forward_only: bool = False,
+                     grad_scaler=None,
+                     dtype: Optional[torch.dtype] = None,
+                     return_losses: bool = False):
+    """Run one 1F1B iteration over ``num_microbatches``.
+
+    Args:
+        fwd_fn: ``fwd_fn(stage_input)`` — for the first stage, stage_input is
+            the sliced micro-batch of ``inputs``; for later stages it is the
+            activation received from the previous stage.  The LAST stage's
+            fwd_fn should return the scalar loss (already averaged over
+            micro-batches or will be summed by the caller).
+        inputs: full-batch input tensor(s), sliced along dim0 (first stage
+            only; other stages may pass None).
+        bwd_fn: optional ``bwd_fn(output, grad_output)`` override.
+        forward_only: run pipelined inference instead (no backward).
+    Returns the last micro-batch's output (last stage) — losses list if
+    ``return_losses``.
+    """
+    pp_rank = tpc.get_pp_rank()
+    pp_size = tpc.get_pp_size()
+    is_first = pp_rank == 0
+    is_last = pp_rank == pp_size - 1
+    bwd = bwd_fn or (lambda out, g: _default_bwd(out, g, optimizer,
+                                                 grad_scaler))
+
+    if pp_size == 1:
+        outs = []
+        for i in range(num_microbatches):
+            out = fwd_fn(_slice_microbatch(inputs, i, num_microbatches))
+            if not forward_only:
+                bwd(out, None)
+            outs.append(out.detach() if torch.is_tensor(out) else out)
+        return outs if return_losses else outs[-1]
+
+    num_warmup = min(pp_size - pp_rank - 1, num_microbatches)
+    num_steady = num_microbatches - num_warmup
+
+    act_shape = None      # shape of activation received from prev stage
+    act_dtype = None
+    out_shape = None      # shape of activation sent to next stage
+    out_dtype = None
+
+    input_store: List[Optional[torch.Tensor]] = []
+    output_store: List[torch.Tensor] = []
+    losses: List[torch.Tensor] = []
+    fwd_idx = 0
+
+    def run_forward(stage_in):
+        nonlocal fwd_idx, out_shape, out_dtype
+        if is_first:
+            stage_in = _slice_microbatch(inputs, fwd_idx, num_microbatches)
+        out = fwd_fn(stage_in)
+        fwd_idx += 1
+        return out
+
+    def recv_act():
+        nonlocal act_shape, act_dtype
+        if is_first:
+            return None
+        if act_shape is None:
+            act_shape, act_dtype = p2p.recv_obj_meta()
+        t = p2p.recv_forward(act_shape, act_dtype)
+        if not forward_only:
+            t.requires_grad_(True)
+        return t
+
+    def send_act(out, first_send: bool):
+        nonlocal out_shape, out_dtype
+        if is_last:
+            if torch.is_tensor(out):
+                losses.append(out.detach())
+            return
+        if first_send:
+            p2p.send_obj_meta(out)
+            out_shape, out_dtype = out.shape, out.dtype
+        p2p.send_forward(out.detach())
+
+    # ---------------- warmup forwards
+    for i in range(num_warmup):
+        stage_in = recv_act()
+        out = run_forward(stage_in)
+        send_act(out, first_send=(i == 0))
+        if not forward_only:
+            input_store.append(stage_in)
+            output_store.append(out)
+
+    # ---------------- steady 1F1B
+    for i in range(num_steady):
+        stage_in = recv_act()
+        out = run_forward(stage_in)
+        if forward_only:
+            send_act(out, first_send=(num_warmup == 0 and i == 0))
+            continue
+        input_store.append(stage_in)
+        output_store.append(out)
+        # fused send-fwd + recv-bwd
+        if is_last:
+            if torch.is_tensor(out):
+                losses.append(out.detach())
+            grad = None
+        else:
+            if num_warmup == 0 and i == 0:
+                p2p.send_obj_meta(out)
+                out_shape, out_dtype = out.shape, out.dtype
+            grad = p2p.send_forward_recv_backward(
+                out.detach(), out.shape, out.dtype)
+        # backward of the OLDEST outstanding micro-batch: grads arrive from
+        # the next stage in micro-batch order, one per steady iteration, so
+        # the received grad always belongs to the oldest outstanding output
+        b_in = input_store.pop(0)
+        b_out = output_store.pop(0)
+        bwd(b_out, grad)
+        if not is_first:
+            g = b_in.grad
+            assert g is not None, "no grad flowed to stage input"
+            # fused send-bwd + recv-fwd happens at next loop's recv; send now
+            p2p.send_backward(g)
+
+    # ---------------- cooldown backwards
+    if not forward_only:
+        for i in range(num_warmup):
+            b_in = input_store.pop(0)
+            b_out = output_store.pop(0)
+            if is_last:
+                grad = None
+            else:
+                grad = p2p.recv_backward(out_shape, out_dtype)
+            bwd(b_out, grad)
+            if not is_first:
+                p2p.send_backward(b_in.grad)
+
+    if return_losses:
+        return losses
+    return losses[-1] if (is_last and losses) else None
+
+
+def forward_eval(fwd_fn: Callable, inputs=None, num_microbatches: int = 1,
+                 gather_outputs: bool = True):
+    """Pipelined inference (reference pipeline_sched.py:233-269): every stage
+    forwards all micro-batches; returns the last stage's outputs list."""
+    pp_rank = tpc.get_pp_rank()
+    pp_size = tpc.get_pp_size()
+    is_first = pp_rank == 0
+    is_last = pp_rank == pp_size - 1
+    outs = []
+    act_shape = act_dtype = None
+    with torch.no_grad():
+        for i in range(num_microbatches):
+            if is_first:
+                stage_in = _slice_microbatch(inputs, i, num_microbatches)
+            else:
+                if act_shape is None:
+                    act_shape, act_dtype = p2p.recv_obj_meta()
+                stage_in = p2p.recv_forward(act_shape, act_dtype)
+            out = fwd_fn(stage_in)
+            if is_last:
+                outs.append(out)
+            else:
+                if i == 0:
+                    p2p.send_obj_meta(out)
+                p2p.send_forward(out)
+    return outs

@@ -5,7 +5,9 @@ from .tp_utils import (set_tp_group, get_tp_group, get_tp_size, get_tp_rank,
                        reduce_scatter_to_sequence_parallel_region,
                        maybe_gather_for_sequence_parallel,
                        maybe_split_into_sequence_parallel,
-                       set_sequence_parallel_attr, is_sequence_parallel)
+                       set_sequence_parallel_attr, is_sequence_parallel,
+                       allreduce_sequence_parallel_grads,
+                       mark_sequence_parallel_params)
 from .mlp import Mlp, TpMlp
 from .attn import Attention, TpAttention
 from .transformer import Block, ParallelBlock, Transformer

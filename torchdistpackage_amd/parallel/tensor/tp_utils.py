@@ -345,3 +345,32 @@ class RowParallelLinear(TpLinear):
         self.weight.copy_(shard)
         if full_bias is not None and self.bias is not None:
             self.bias.copy_(full_bias)
+
+
+def mark_sequence_parallel_params(module: "nn.Module"):
+    """Tag a module's params as SP-region params (grads computed from the
+    local sequence shard only -> need an all-reduce over TP)."""
+    for p in module.parameters():
+        p.sequence_parallel_param = True
+
+
+def allreduce_sequence_parallel_grads(module: "nn.Module"):
+    """All-reduce the grads of SP-region params (LayerNorm weights/biases
+    inside ParallelBlock) over the TP group.  Must be called once per
+    iteration after backward when sequence_parallel is on — Megatron-SP
+    semantics the reference does not implement (its transformer tolerates
+    rtol=1e-1, test_transformer.py:33-39)."""
+    if get_tp_size() == 1:
+        return
+    group = get_tp_group()
+    grads = [p.grad for p in module.parameters()
+             if getattr(p, "sequence_parallel_param", False)
+             and p.grad is not None]
+    if not grads:
+        return
+    flat = torch.cat([g.reshape(-1) for g in grads])
+    dist.all_reduce(flat, group=group)
+    off = 0
+    for g in grads:
+        g.copy_(flat[off:off + g.numel()].view_as(g))
+        off += g.numel()

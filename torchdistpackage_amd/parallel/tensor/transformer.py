@@ -64,6 +64,12 @@ class ParallelBlock(nn.Module):
         self.ln_2 = LayerNorm(dim, **kw)
         self.mlp = TpMlp(dim, hidden_mult, bias=bias,
                          sequence_parallel=self.sequence_parallel, **kw)
+        if self.sequence_parallel:
+            # LN grads come from the local sequence shard -> all-reduce over
+            # TP needed once per iteration (allreduce_sequence_parallel_grads)
+            from .tp_utils import mark_sequence_parallel_params
+            mark_sequence_parallel_params(self.ln_1)
+            mark_sequence_parallel_params(self.ln_2)
 
     def forward(self, x):
         if self.sequence_parallel:
