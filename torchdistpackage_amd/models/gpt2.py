@@ -110,7 +110,9 @@ class GPT2Model(nn.Module):
         from ..parallel.tensor import (is_sequence_parallel,
                                        gather_from_sequence_parallel_region)
         if is_sequence_parallel(x) and get_tp_size() > 1:
-            x = gather_from_sequence_parallel_region(x)
+            # split-backward: the head + loss are computed identically on
+            # every TP rank, so each rank's grad is already complete
+            x = gather_from_sequence_parallel_region(x, bwd_mode="split")
         logits = self.head(x)
         out = {"logits": logits}
         if labels is not None:

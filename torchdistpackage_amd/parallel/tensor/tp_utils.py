@@ -96,6 +96,9 @@ def _reduce_scatter_first_dim(x: torch.Tensor) -> torch.Tensor:
     out_shape[0] //= tp
     x = x.contiguous()
     if _backend_is_gloo(group):
+        # clone: all_reduce mutates in place, and backward inputs (incoming
+        # grads) must never be mutated — they may be shared by other consumers
+        x = x.clone()
         dist.all_reduce(x, group=group)
         r = dist.get_rank(group)
         return x.narrow(0, r * out_shape[0], out_shape[0]).clone()
@@ -145,7 +148,8 @@ class _CopyToTp(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad):
-        return _all_reduce(grad.contiguous())
+        # clone: never mutate the incoming grad in place (all_reduce would)
+        return _all_reduce(grad.contiguous().clone())
 
 
 class _ReduceScatterToSp(torch.autograd.Function):

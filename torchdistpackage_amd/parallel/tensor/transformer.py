@@ -113,5 +113,10 @@ class Transformer(nn.Module):
         for blk in self.blocks:
             x = blk(x)
         if is_sequence_parallel(x) and get_tp_size() > 1:
-            x = gather_from_sequence_parallel_region(x)
+            # final gather feeds replicated compute (loss/head computed
+            # identically on every TP rank): each rank's backward grad is
+            # already the full gradient, so backward takes the local slice
+            # (reference tp_utils.py:126-149 'already summed' mode) — NOT
+            # reduce-scatter, which would double-count
+            x = gather_from_sequence_parallel_region(x, bwd_mode="split")
         return x
