@@ -108,7 +108,12 @@ class NaiveDdp(nn.Module):
         # every num_grad_acc_iter-th fire communicates.
         self._fires: Dict[int, int] = {}
 
-        self._params = [p for p in module.parameters() if p.requires_grad]
+        # expert-parallel params are excluded: they are different per EP rank
+        # and sync over 'moe_dp' via MoEDP instead (reference moe_dp.md)
+        def _is_expert(p):
+            return getattr(p, "expert_parallel", False)
+        self._params = [p for p in module.parameters()
+                        if p.requires_grad and not _is_expert(p)]
         self._param_bucket: Dict[int, tuple] = {}  # id(p) -> (bucket, idx)
         self._buckets: List[GradBucket] = []
         self._hooks = []
@@ -122,6 +127,8 @@ class NaiveDdp(nn.Module):
                 dist.get_world_size(self.group) > 1:
             with torch.no_grad():
                 for p in module.parameters():
+                    if _is_expert(p):
+                        continue
                     dist.broadcast(p.data, src=self._group_src(), group=self.group)
                 for b in module.buffers():
                     if b.dtype.is_floating_point or b.dtype in (
