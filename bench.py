@@ -70,7 +70,10 @@ def main():
             torch.cuda.set_device(0)
     import torch.distributed as dist
 
-    dp, pp, tp = PARALLEL_MAP.get(world, (world, 1, 1))
+    if os.environ.get("TDPA_PARALLEL"):   # "dp,pp,tp" override for testing
+        dp, pp, tp = (int(v) for v in os.environ["TDPA_PARALLEL"].split(","))
+    else:
+        dp, pp, tp = PARALLEL_MAP.get(world, (world, 1, 1))
     if world > 1:
         tpc.setup_process_groups(
             [("data", dp), ("pipe", pp), ("tensor", tp)])
