@@ -1,22 +1,27 @@
-// Flash-attention forward for gfx950 (CDNA4 MFMA, online softmax).
+// Flash-attention forward + backward for gfx950 (CDNA4 MFMA, online softmax).
 //
 // Blockwise algorithm per the reference's in-repo spec
-// (/root/reference/explore/flash-attn/tile_attn.py:100-154): per Q-tile,
-// iterate K/V tiles maintaining running row max m and exp-sum l; output is
-// rescaled by exp(m_old - m_new) at each tile; LSE = m + log(l) is saved for
-// the backward.
+// (/root/reference/explore/flash-attn/tile_attn.py:100-212): per Q-tile,
+// iterate K/V tiles maintaining running row max m and exp-sum l; LSE is saved
+// for the backward; backward = two recompute passes (dq; dk+dv) plus a
+// delta = rowsum(do*o) preprocess.  No atomics anywhere.
 //
-// v0 structure (correctness-first; optimization ladder applied in-place later):
-//   - workgroup = 256 threads = 4 waves; each wave owns 16 q-rows, the block
-//     owns a 64-row Q tile of one (batch, head)
-//   - K/V tiles of 32 keys staged in LDS (V stored transposed [D][32] so the
-//     PV B-fragment reads are contiguous ds_read_b128)
-//   - QK^T and PV on v_mfma_f32_16x16x32_bf16 (A/B: 8 bf16/lane, K-contig
-//     per lane; C/D: col=lane&15, row=(lane>>4)*4+reg)
-//   - P is round-tripped through LDS to convert C-layout -> A-layout
-//   - f32 accumulation throughout; bf16 only at memory boundaries
+// v1 structure (v0 + the two biggest levers from the CDNA4 guide):
+//   - LDS XOR swizzles on every MFMA B-fragment tile: without them a
+//     ds_read_b128 of 16 rows at one column slot is up to 16-way
+//     bank-conflicted (guide 6.4: this one conflict was half an attention
+//     kernel's time).  256B-row tiles use col ^= (row & (D/8-1)) << 3,
+//     32-short-row tiles use col ^= ((row>>2) & 3) << 3 (short-index space).
+//   - Strided global addressing: q/k/v/o/do/dq/dk/dv are (B, H, S, D) VIEWS
+//     with arbitrary b/h/s strides (last dim contiguous).  The training path
+//     passes views straight into the fused (S, B, 3*H*D) qkv buffer, so no
+//     permute-contiguous copies happen anywhere around attention.
 //
-// Supported: head_dim 64 / 128, any S (K-tail masked), causal or full.
+// Workgroup = 256 threads = 4 waves; wave owns 16 q-rows (fwd/dq) or 16 keys
+// (dkdv); K/V tiles of 32 staged in LDS; v_mfma_f32_16x16x32_bf16
+// (A/B: 8 bf16/lane K-contiguous; C/D: col=lane&15, row=(lane>>4)*4+reg —
+// layout verified on hardware by tests/test_ops_gpu.py::test_mfma_layout_probe).
+// Supported: head_dim 64 / 128, any S, causal or full.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -27,15 +32,31 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-constexpr int QT = 64;    // q rows per workgroup
-constexpr int WQ = 16;    // q rows per wave
-constexpr int KT = 32;    // keys per kv tile
+constexpr int QT = 64;    // q rows (or keys, in dkdv) per workgroup
+constexpr int WQ = 16;    // rows per wave
+constexpr int KT = 32;    // keys (or q rows, in dkdv) per LDS tile
 constexpr int NWAVE = 4;
 
+struct Strides {            // element strides of a (B,H,S,D) view
+  long b, h, s;
+};
+
 DEVINL bf16x8_v pack8(const unsigned short* p) {
-  // reinterpret 8 contiguous bf16 (16 B, must be 16B-aligned in LDS/global)
   return *(const bf16x8_v*)p;
 }
+
+// swizzled index (in shorts) into a [rows][D] tile with 2*D-byte rows
+template <int D>
+DEVINL int swzD(int row, int col) {
+  return row * D + (col ^ ((row & (D / 8 - 1)) << 3));
+}
+
+// swizzled index into a [rows][KT=32] tile (64-byte rows)
+DEVINL int swz32(int row, int col) {
+  return row * KT + (col ^ (((row >> 2) & 3) << 3));
+}
+
+// ---------------------------------------------------------------- forward
 
 template <int D, bool CAUSAL>
 __launch_bounds__(256)
@@ -44,41 +65,38 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
                                 const unsigned short* __restrict__ v,
                                 unsigned short* __restrict__ o,
                                 float* __restrict__ lse,
+                                Strides qs, Strides ks, Strides vs, Strides os,
                                 int B, int H, int S, float scale) {
-  constexpr int KC = D / 32;     // QK^T k-chunks
-  constexpr int DC = D / 16;     // PV d-chunks (output col tiles)
-  // LDS: K [KT][D] bf16, V^T [D][KT] bf16, P [NWAVE][16][KT+pad?] bf16
-  __shared__ unsigned short k_lds[KT][D];
-  __shared__ unsigned short vt_lds[D][KT];
-  __shared__ unsigned short p_lds[NWAVE][16][KT];
+  constexpr int KC = D / 32;
+  constexpr int DC = D / 16;
+  __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];        // K[key][d], swzD
+  __shared__ __attribute__((aligned(16))) unsigned short vt_lds[D * KT];       // V^T[d][key], swz32
+  __shared__ __attribute__((aligned(16))) unsigned short p_lds[NWAVE * 16 * KT];  // P, swz32 per wave
 
-  const int bh = blockIdx.y;          // batch*H + head
-  const int qtile = blockIdx.x;
-  const int qbase = qtile * QT;
+  const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
+  const int qbase = blockIdx.x * QT;
   if (qbase >= S) return;
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;
   const int lane = tid % WAVE;
   const int l15 = lane & 15;
-  const int lg = lane >> 4;           // 16-lane group id (0..3)
+  const int lg = lane >> 4;
 
-  const long bh_off = (long)bh * S * D;
-  const unsigned short* qp = q + bh_off;
-  const unsigned short* kp = k + bh_off;
-  const unsigned short* vp = v + bh_off;
+  const unsigned short* qp = q + bb * qs.b + hh * qs.h;
+  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
 
-  // ---- load Q fragments: wave w covers rows qbase + w*16 + (0..15)
   const int qrow0 = qbase + wid * WQ;
   bf16x8_v a_q[KC];
   {
     int r = qrow0 + l15;
-    int rr = r < S ? r : S - 1;       // clamp; invalid rows never stored
+    long rr = (r < S ? r : S - 1) * qs.s;
 #pragma unroll
     for (int c = 0; c < KC; ++c)
-      a_q[c] = pack8(qp + (long)rr * D + c * 32 + lg * 8);
+      a_q[c] = pack8(qp + rr + c * 32 + lg * 8);
   }
 
-  // ---- running state: 4 rows per lane (rows lg*4+rr of this wave's 16)
   float m_run[4], l_run[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) { m_run[i] = -1e30f; l_run[i] = 0.f; }
@@ -89,49 +107,44 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
   const int kv_end = CAUSAL ? min(S, qbase + QT) : S;
 
   for (int kt0 = 0; kt0 < kv_end; kt0 += KT) {
-    // ---- stage K tile [KT][D] and V^T tile [D][KT]
     __syncthreads();
     {
-      // 256 threads load KT*D elements; 8 bf16 per thread-step
       const int elems = KT * D;
       for (int idx = tid * 8; idx < elems; idx += 256 * 8) {
         int key = idx / D;
         int col = idx % D;
         int gkey = kt0 + key;
         if (gkey < S) {
-          bf16x8_v kv8 = pack8(kp + (long)gkey * D + col);
-          *(bf16x8_v*)&k_lds[key][col] = kv8;
-          bf16x8_v vv8 = pack8(vp + (long)gkey * D + col);
-          // transpose-store V: vt[col + j][key]
+          bf16x8_v kv8 = pack8(kp + (long)gkey * ks.s + col);
+          *(bf16x8_v*)&k_lds[swzD<D>(key, col)] = kv8;
+          bf16x8_v vv8 = pack8(vp + (long)gkey * vs.s + col);
           const unsigned short* vsrc = (const unsigned short*)&vv8;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) vt_lds[col + j][key] = vsrc[j];
+          for (int j = 0; j < 8; ++j)
+            vt_lds[swz32(col + j, key)] = vsrc[j];
         } else {
-          // zero-fill tail (scores masked anyway, V contributes 0)
           for (int j = 0; j < 8; ++j) {
-            k_lds[key][col + j] = 0;
-            vt_lds[col + j][key] = 0;
+            k_lds[swzD<D>(key, col + j - (col + j) % 8) +
+                  ((col + j) % 8)] = 0;  // keep swizzle chunk-aligned
+            vt_lds[swz32(col + j, key)] = 0;
           }
         }
       }
     }
     __syncthreads();
 
-    // ---- S = scale * Q K^T for this wave's 16 rows x KT keys
     f32x4 s_acc[KT / 16];
 #pragma unroll
     for (int kg = 0; kg < KT / 16; ++kg) {
       s_acc[kg] = (f32x4)(0.f);
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
-        bf16x8_v b_k = pack8(&k_lds[kg * 16 + l15][c * 32 + lg * 8]);
+        bf16x8_v b_k = pack8(&k_lds[swzD<D>(kg * 16 + l15, c * 32 + lg * 8)]);
         s_acc[kg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_q[c], b_k, s_acc[kg], 0, 0, 0);
       }
     }
 
-    // ---- mask + online softmax update
-    // C layout: value (row = lg*4 + rr, col = l15) per reg rr
     float p_val[KT / 16][4];
     float alpha[4];
     {
@@ -169,34 +182,29 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
       }
     }
 
-    // ---- rescale O accumulators by alpha (acc rows = lg*4+rr)
 #pragma unroll
     for (int d = 0; d < DC; ++d)
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) acc_o[d][rr] *= alpha[rr];
 
-    // ---- P -> LDS (C layout -> A layout via memory)
+    unsigned short* pw = &p_lds[wid * 16 * KT];
 #pragma unroll
     for (int kg = 0; kg < KT / 16; ++kg)
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr)
-        p_lds[wid][lg * 4 + rr][kg * 16 + l15] = f2bf(p_val[kg][rr]);
-    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): LDS writes visible in-wave
-    // (wave-private p_lds slice: no cross-wave barrier needed)
+        pw[swz32(lg * 4 + rr, kg * 16 + l15)] = f2bf(p_val[kg][rr]);
 
-    // ---- O += P V  (A = P [16][KT], B = V [KT][16-col chunk])
-    bf16x8_v a_p = pack8(&p_lds[wid][l15][lg * 8]);
+    bf16x8_v a_p = pack8(&pw[swz32(l15, lg * 8)]);
 #pragma unroll
     for (int d = 0; d < DC; ++d) {
-      bf16x8_v b_v = pack8(&vt_lds[d * 16 + l15][lg * 8]);
+      bf16x8_v b_v = pack8(&vt_lds[swz32(d * 16 + l15, lg * 8)]);
       acc_o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           a_p, b_v, acc_o[d], 0, 0, 0);
     }
   }
 
-  // ---- epilogue: O /= l, store bf16; LSE = m + log(l)
-  unsigned short* op = o + bh_off;
-  float* lsep = lse + (long)bh * S;
+  unsigned short* op = o + bb * os.b + hh * os.h;
+  float* lsep = lse + ((long)bh) * S;
 #pragma unroll
   for (int rr = 0; rr < 4; ++rr) {
     int qrow = qrow0 + lg * 4 + rr;
@@ -204,66 +212,28 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
     float inv_l = l_run[rr] > 0.f ? 1.f / l_run[rr] : 0.f;
 #pragma unroll
     for (int d = 0; d < DC; ++d)
-      op[(long)qrow * D + d * 16 + l15] = f2bf(acc_o[d][rr] * inv_l);
+      op[(long)qrow * os.s + d * 16 + l15] = f2bf(acc_o[d][rr] * inv_l);
     if (l15 == 0)
       lsep[qrow] = m_run[rr] + __logf(l_run[rr] > 0.f ? l_run[rr] : 1.f);
   }
 }
 
-}  // namespace
-
-std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, bool causal,
-                                    double scale) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
-              "attn_fwd: bf16 CUDA tensors required");
-  TORCH_CHECK(q.dim() == 4, "q must be (B,H,S,D)");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
-  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
-  TORCH_CHECK(k.size(2) == S, "cross-attention S_kv != S_q not supported yet");
-  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
-  auto o = torch::empty_like(q);
-  auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
-  auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid((S + QT - 1) / QT, B * H), block(256);
-  const unsigned short* qp = (const unsigned short*)q.data_ptr();
-  const unsigned short* kp = (const unsigned short*)k.data_ptr();
-  const unsigned short* vp = (const unsigned short*)v.data_ptr();
-  unsigned short* op = (unsigned short*)o.data_ptr();
-  float* lp = lse.data_ptr<float>();
-#define LAUNCH(DD, CC)                                                        \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, block, 0, stream, qp,   \
-                     kp, vp, op, lp, B, H, S, (float)scale)
-  if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
-  else          { if (causal) LAUNCH(64, true);  else LAUNCH(64, false);  }
-#undef LAUNCH
-  HIP_CHECK_LAST();
-  return {o, lse};
-}
-
-// ===========================================================================
-// Flash-attention backward (FA2-style two recompute passes, reference math
-// spec: /root/reference/explore/flash-attn/tile_attn.py:156-212).
-//   delta kernel: delta = rowsum(do * o)               (memory-bound)
-//   dq kernel:    per Q-tile, loop KV tiles: recompute P from LSE,
-//                 dp = do V^T, ds = P (dp - delta) scale, dq += ds K
-//   dkdv kernel:  per KV-tile, loop Q tiles: recompute P^T,
-//                 dv += P^T do, ds^T = P^T (dp^T - delta) scale, dk += ds^T q
-// No atomics: each output row is owned by exactly one workgroup.
-// ===========================================================================
-
-namespace {
+// ---------------------------------------------------------------- backward
 
 __global__ void attn_delta_kernel(const unsigned short* __restrict__ o,
                                   const unsigned short* __restrict__ dout,
                                   float* __restrict__ delta,
-                                  long rows, int D) {
-  // one wave per row; vectorized 8-wide
+                                  Strides os, Strides ds,
+                                  int B, int H, int S, int D) {
+  // one wave per row (grid.x covers B*H*S rows), vectorized 8-wide
   long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
-  if (row >= rows) return;
+  if (row >= (long)B * H * S) return;
   const int lane = threadIdx.x % WAVE;
-  const unsigned short* orow = o + row * D;
-  const unsigned short* drow = dout + row * D;
+  int s = row % S;
+  int h = (row / S) % H;
+  int b = row / ((long)S * H);
+  const unsigned short* orow = o + b * os.b + h * os.h + (long)s * os.s;
+  const unsigned short* drow = dout + b * ds.b + h * ds.h + (long)s * ds.s;
   float acc = 0.f;
   for (int i = lane * 8; i < D; i += WAVE * 8) {
     bf16x8_v ov = pack8(orow + i);
@@ -286,17 +256,19 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                                    const float* __restrict__ lse,
                                    const float* __restrict__ delta,
                                    unsigned short* __restrict__ dq,
+                                   Strides qs, Strides ks, Strides vs,
+                                   Strides dos, Strides dqs,
                                    int B, int H, int S, float scale) {
   constexpr int KC = D / 32;
   constexpr int DC = D / 16;
-  __shared__ unsigned short k_lds[KT][D];     // K[key][d] for QK^T
-  __shared__ unsigned short kt_lds[D][KT];    // K^T[d][key] for ds@K
-  __shared__ unsigned short v_lds[KT][D];     // V[key][d] for do V^T
-  __shared__ unsigned short ds_lds[NWAVE][16][KT];
+  __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];     // K[key][d], swzD
+  __shared__ __attribute__((aligned(16))) unsigned short kt_lds[D * KT];    // K^T[d][key], swz32
+  __shared__ __attribute__((aligned(16))) unsigned short v_lds[KT * D];     // V[key][d], swzD
+  __shared__ __attribute__((aligned(16))) unsigned short ds_lds[NWAVE * 16 * KT];  // dS, swz32
 
   const int bh = blockIdx.y;
-  const int qtile = blockIdx.x;
-  const int qbase = qtile * QT;
+  const int bb = bh / H, hh = bh % H;
+  const int qbase = blockIdx.x * QT;
   if (qbase >= S) return;
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;
@@ -304,11 +276,10 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
-  const long bh_off = (long)bh * S * D;
-  const unsigned short* qp = q + bh_off;
-  const unsigned short* kp = k + bh_off;
-  const unsigned short* vp = v + bh_off;
-  const unsigned short* dop = dout + bh_off;
+  const unsigned short* qp = q + bb * qs.b + hh * qs.h;
+  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
+  const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
   const float* lsep = lse + (long)bh * S;
   const float* delp = delta + (long)bh * S;
 
@@ -317,11 +288,11 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   float my_lse[4], my_delta[4];
   {
     int r = qrow0 + l15;
-    int rr = r < S ? r : S - 1;
+    int rc = r < S ? r : S - 1;
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
-      a_q[c] = pack8(qp + (long)rr * D + c * 32 + lg * 8);
-      a_do[c] = pack8(dop + (long)rr * D + c * 32 + lg * 8);
+      a_q[c] = pack8(qp + (long)rc * qs.s + c * 32 + lg * 8);
+      a_do[c] = pack8(dop + (long)rc * dos.s + c * 32 + lg * 8);
     }
 #pragma unroll
     for (int rr4 = 0; rr4 < 4; ++rr4) {
@@ -346,25 +317,25 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
         int col = idx % D;
         int gkey = kt0 + key;
         if (gkey < S) {
-          bf16x8_v kv8 = pack8(kp + (long)gkey * D + col);
-          *(bf16x8_v*)&k_lds[key][col] = kv8;
-          bf16x8_v vv8 = pack8(vp + (long)gkey * D + col);
-          *(bf16x8_v*)&v_lds[key][col] = vv8;
+          bf16x8_v kv8 = pack8(kp + (long)gkey * ks.s + col);
+          *(bf16x8_v*)&k_lds[swzD<D>(key, col)] = kv8;
+          bf16x8_v vv8 = pack8(vp + (long)gkey * vs.s + col);
+          *(bf16x8_v*)&v_lds[swzD<D>(key, col)] = vv8;
           const unsigned short* ksrc = (const unsigned short*)&kv8;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) kt_lds[col + j][key] = ksrc[j];
+          for (int j = 0; j < 8; ++j)
+            kt_lds[swz32(col + j, key)] = ksrc[j];
         } else {
           for (int j = 0; j < 8; ++j) {
-            k_lds[key][col + j] = 0;
-            v_lds[key][col + j] = 0;
-            kt_lds[col + j][key] = 0;
+            k_lds[swzD<D>(key, col) + j] = 0;
+            v_lds[swzD<D>(key, col) + j] = 0;
+            kt_lds[swz32(col + j, key)] = 0;
           }
         }
       }
     }
     __syncthreads();
 
-    // S and dP tiles for 16 q-rows x KT keys
     f32x4 s_acc[KT / 16], dp_acc[KT / 16];
 #pragma unroll
     for (int kg = 0; kg < KT / 16; ++kg) {
@@ -372,16 +343,16 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
       dp_acc[kg] = (f32x4)(0.f);
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
-        bf16x8_v b_k = pack8(&k_lds[kg * 16 + l15][c * 32 + lg * 8]);
+        bf16x8_v b_k = pack8(&k_lds[swzD<D>(kg * 16 + l15, c * 32 + lg * 8)]);
         s_acc[kg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_q[c], b_k, s_acc[kg], 0, 0, 0);
-        bf16x8_v b_v = pack8(&v_lds[kg * 16 + l15][c * 32 + lg * 8]);
+        bf16x8_v b_v = pack8(&v_lds[swzD<D>(kg * 16 + l15, c * 32 + lg * 8)]);
         dp_acc[kg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_do[c], b_v, dp_acc[kg], 0, 0, 0);
       }
     }
 
-    // ds = P * (dP - delta) * scale, P = exp(S*scale - lse)
+    unsigned short* dsw = &ds_lds[wid * 16 * KT];
 #pragma unroll
     for (int kg = 0; kg < KT / 16; ++kg) {
 #pragma unroll
@@ -392,28 +363,27 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
         float p = valid ?
             __expf(s_acc[kg][rr] * scale - my_lse[rr]) : 0.f;
         float ds = p * (dp_acc[kg][rr] - my_delta[rr]) * scale;
-        ds_lds[wid][lg * 4 + rr][kg * 16 + l15] = f2bf(ds);
+        dsw[swz32(lg * 4 + rr, kg * 16 + l15)] = f2bf(ds);
       }
     }
 
-    // dq += ds @ K   (A = ds [16][KT], B = K [KT][16 d-cols])
-    bf16x8_v a_ds = pack8(&ds_lds[wid][l15][lg * 8]);
+    bf16x8_v a_ds = pack8(&dsw[swz32(l15, lg * 8)]);
 #pragma unroll
     for (int d = 0; d < DC; ++d) {
-      bf16x8_v b_kt = pack8(&kt_lds[d * 16 + l15][lg * 8]);
+      bf16x8_v b_kt = pack8(&kt_lds[swz32(d * 16 + l15, lg * 8)]);
       acc_dq[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           a_ds, b_kt, acc_dq[d], 0, 0, 0);
     }
   }
 
-  unsigned short* dqp = dq + bh_off;
+  unsigned short* dqp = dq + bb * dqs.b + hh * dqs.h;
 #pragma unroll
   for (int rr = 0; rr < 4; ++rr) {
     int qrow = qrow0 + lg * 4 + rr;
     if (qrow >= S) continue;
 #pragma unroll
     for (int d = 0; d < DC; ++d)
-      dqp[(long)qrow * D + d * 16 + l15] = f2bf(acc_dq[d][rr]);
+      dqp[(long)qrow * dqs.s + d * 16 + l15] = f2bf(acc_dq[d][rr]);
   }
 }
 
@@ -427,22 +397,22 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
                                      const float* __restrict__ delta,
                                      unsigned short* __restrict__ dk,
                                      unsigned short* __restrict__ dv,
+                                     Strides qs, Strides ks, Strides vs,
+                                     Strides dos, Strides dks, Strides dvs,
                                      int B, int H, int S, float scale) {
   constexpr int KC = D / 32;
   constexpr int DC = D / 16;
-  // workgroup owns 64 keys (wave w: keys ktile*64 + w*16 + 0..15);
-  // loops over q tiles of 32 rows
-  __shared__ unsigned short q_lds[KT][D];     // Q[qrow][d]   (KT=32 q rows)
-  __shared__ unsigned short qt_lds[D][KT];    // Q^T[d][qrow]
-  __shared__ unsigned short do_lds[KT][D];    // dO[qrow][d]
-  __shared__ unsigned short dot_lds[D][KT];   // dO^T[d][qrow]
-  __shared__ unsigned short st_lds[NWAVE][16][KT];  // P^T / dS^T staging
-  __shared__ float lse_lds[KT];
-  __shared__ float del_lds[KT];
+  __shared__ __attribute__((aligned(16))) unsigned short q_lds[KT * D];     // Q[qrow][d], swzD
+  __shared__ __attribute__((aligned(16))) unsigned short qt_lds[D * KT];    // Q^T[d][qrow], swz32
+  __shared__ __attribute__((aligned(16))) unsigned short do_lds[KT * D];    // dO[qrow][d], swzD
+  __shared__ __attribute__((aligned(16))) unsigned short dot_lds[D * KT];   // dO^T[d][qrow], swz32
+  __shared__ __attribute__((aligned(16))) unsigned short st_lds[NWAVE * 16 * KT];  // P^T / dS^T, swz32
+  __shared__ __attribute__((aligned(16))) float lse_lds[KT];
+  __shared__ __attribute__((aligned(16))) float del_lds[KT];
 
   const int bh = blockIdx.y;
-  const int ktile = blockIdx.x;
-  const int kbase = ktile * QT;   // 64 keys per workgroup
+  const int bb = bh / H, hh = bh % H;
+  const int kbase = blockIdx.x * QT;
   if (kbase >= S) return;
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;
@@ -450,24 +420,22 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
-  const long bh_off = (long)bh * S * D;
-  const unsigned short* qp = q + bh_off;
-  const unsigned short* kp = k + bh_off;
-  const unsigned short* vp = v + bh_off;
-  const unsigned short* dop = dout + bh_off;
+  const unsigned short* qp = q + bb * qs.b + hh * qs.h;
+  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
+  const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
   const float* lsep = lse + (long)bh * S;
   const float* delp = delta + (long)bh * S;
 
-  // this wave's 16 keys: fragments of K and V (A-operand layout)
   const int key0 = kbase + wid * WQ;
   bf16x8_v a_k[KC], a_v[KC];
   {
     int r = key0 + l15;
-    int rr = r < S ? r : S - 1;
+    long rc = (long)(r < S ? r : S - 1);
 #pragma unroll
     for (int c = 0; c < KC; ++c) {
-      a_k[c] = pack8(kp + (long)rr * D + c * 32 + lg * 8);
-      a_v[c] = pack8(vp + (long)rr * D + c * 32 + lg * 8);
+      a_k[c] = pack8(kp + rc * ks.s + c * 32 + lg * 8);
+      a_v[c] = pack8(vp + rc * vs.s + c * 32 + lg * 8);
     }
   }
 
@@ -478,7 +446,7 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
     acc_dv[d] = (f32x4)(0.f);
   }
 
-  const int qt_start = CAUSAL ? (kbase / KT) * KT : 0;
+  const int qt_start = CAUSAL ? kbase : 0;
 
   for (int qt0 = qt_start; qt0 < S; qt0 += KT) {
     __syncthreads();
@@ -489,23 +457,23 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
         int col = idx % D;
         int grow = qt0 + row;
         if (grow < S) {
-          bf16x8_v qv8 = pack8(qp + (long)grow * D + col);
-          *(bf16x8_v*)&q_lds[row][col] = qv8;
-          bf16x8_v dv8 = pack8(dop + (long)grow * D + col);
-          *(bf16x8_v*)&do_lds[row][col] = dv8;
-          const unsigned short* qs = (const unsigned short*)&qv8;
-          const unsigned short* ds = (const unsigned short*)&dv8;
+          bf16x8_v qv8 = pack8(qp + (long)grow * qs.s + col);
+          *(bf16x8_v*)&q_lds[swzD<D>(row, col)] = qv8;
+          bf16x8_v dv8 = pack8(dop + (long)grow * dos.s + col);
+          *(bf16x8_v*)&do_lds[swzD<D>(row, col)] = dv8;
+          const unsigned short* qsrc = (const unsigned short*)&qv8;
+          const unsigned short* dsrc = (const unsigned short*)&dv8;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            qt_lds[col + j][row] = qs[j];
-            dot_lds[col + j][row] = ds[j];
+            qt_lds[swz32(col + j, row)] = qsrc[j];
+            dot_lds[swz32(col + j, row)] = dsrc[j];
           }
         } else {
           for (int j = 0; j < 8; ++j) {
-            q_lds[row][col + j] = 0;
-            do_lds[row][col + j] = 0;
-            qt_lds[col + j][row] = 0;
-            dot_lds[col + j][row] = 0;
+            q_lds[swzD<D>(row, col) + j] = 0;
+            do_lds[swzD<D>(row, col) + j] = 0;
+            qt_lds[swz32(col + j, row)] = 0;
+            dot_lds[swz32(col + j, row)] = 0;
           }
         }
       }
@@ -517,7 +485,6 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
     }
     __syncthreads();
 
-    // S^T and dP^T tiles: 16 keys x KT q-rows
     f32x4 st_acc[KT / 16], dpt_acc[KT / 16];
 #pragma unroll
     for (int qg = 0; qg < KT / 16; ++qg) {
@@ -525,118 +492,150 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
       dpt_acc[qg] = (f32x4)(0.f);
 #pragma unroll
       for (int c = 0; c < KC; ++c) {
-        bf16x8_v b_q = pack8(&q_lds[qg * 16 + l15][c * 32 + lg * 8]);
+        bf16x8_v b_q = pack8(&q_lds[swzD<D>(qg * 16 + l15, c * 32 + lg * 8)]);
         st_acc[qg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_k[c], b_q, st_acc[qg], 0, 0, 0);
-        bf16x8_v b_do = pack8(&do_lds[qg * 16 + l15][c * 32 + lg * 8]);
+        bf16x8_v b_do = pack8(&do_lds[swzD<D>(qg * 16 + l15, c * 32 + lg * 8)]);
         dpt_acc[qg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a_v[c], b_do, dpt_acc[qg], 0, 0, 0);
       }
     }
 
-    // P^T = exp(S^T*scale - lse[qcol]); stage P^T for the dv MFMA
+    unsigned short* stw = &st_lds[wid * 16 * KT];
 #pragma unroll
     for (int qg = 0; qg < KT / 16; ++qg) {
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        int key = key0 + lg * 4 + rr;     // C-layout row = key here
-        int qrow = qt0 + qg * 16 + l15;   // C-layout col = q
+        int key = key0 + lg * 4 + rr;
+        int qrow = qt0 + qg * 16 + l15;
         bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
         float p = valid ?
             __expf(st_acc[qg][rr] * scale - lse_lds[qg * 16 + l15]) : 0.f;
-        st_acc[qg][rr] = p;   // reuse as P^T
-        st_lds[wid][lg * 4 + rr][qg * 16 + l15] = f2bf(p);
+        st_acc[qg][rr] = p;
+        stw[swz32(lg * 4 + rr, qg * 16 + l15)] = f2bf(p);
       }
     }
 
-    // dv += P^T @ dO  (A = P^T [16keys][KT q], B = dO [q][16 d-cols])
-    bf16x8_v a_pt = pack8(&st_lds[wid][l15][lg * 8]);
+    bf16x8_v a_pt = pack8(&stw[swz32(l15, lg * 8)]);
 #pragma unroll
     for (int d = 0; d < DC; ++d) {
-      bf16x8_v b_dot = pack8(&dot_lds[d * 16 + l15][lg * 8]);
+      bf16x8_v b_dot = pack8(&dot_lds[swz32(d * 16 + l15, lg * 8)]);
       acc_dv[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           a_pt, b_dot, acc_dv[d], 0, 0, 0);
     }
 
-    // dS^T = P^T * (dP^T - delta[qcol]) * scale; restage
-    __syncthreads();  // st_lds reuse: everyone done reading P^T
 #pragma unroll
     for (int qg = 0; qg < KT / 16; ++qg) {
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         float ds = st_acc[qg][rr] *
             (dpt_acc[qg][rr] - del_lds[qg * 16 + l15]) * scale;
-        st_lds[wid][lg * 4 + rr][qg * 16 + l15] = f2bf(ds);
+        stw[swz32(lg * 4 + rr, qg * 16 + l15)] = f2bf(ds);
       }
     }
 
-    // dk += dS^T @ Q  (A = dS^T [16keys][KT q], B = Q [q][16 d-cols])
-    bf16x8_v a_dst = pack8(&st_lds[wid][l15][lg * 8]);
+    bf16x8_v a_dst = pack8(&stw[swz32(l15, lg * 8)]);
 #pragma unroll
     for (int d = 0; d < DC; ++d) {
-      bf16x8_v b_qt = pack8(&qt_lds[d * 16 + l15][lg * 8]);
+      bf16x8_v b_qt = pack8(&qt_lds[swz32(d * 16 + l15, lg * 8)]);
       acc_dk[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           a_dst, b_qt, acc_dk[d], 0, 0, 0);
     }
   }
 
-  unsigned short* dkp = dk + bh_off;
-  unsigned short* dvp = dv + bh_off;
+  unsigned short* dkp = dk + bb * dks.b + hh * dks.h;
+  unsigned short* dvp = dv + bb * dvs.b + hh * dvs.h;
 #pragma unroll
   for (int rr = 0; rr < 4; ++rr) {
     int key = key0 + lg * 4 + rr;
     if (key >= S) continue;
 #pragma unroll
     for (int d = 0; d < DC; ++d) {
-      dkp[(long)key * D + d * 16 + l15] = f2bf(acc_dk[d][rr]);
-      dvp[(long)key * D + d * 16 + l15] = f2bf(acc_dv[d][rr]);
+      dkp[(long)key * dks.s + d * 16 + l15] = f2bf(acc_dk[d][rr]);
+      dvp[(long)key * dvs.s + d * 16 + l15] = f2bf(acc_dv[d][rr]);
     }
   }
 }
 
+Strides get_strides(const torch::Tensor& t) {
+  TORCH_CHECK(t.dim() == 4 && t.stride(3) == 1,
+              "attention tensors must be 4D (B,H,S,D) with contiguous D");
+  return Strides{t.stride(0), t.stride(1), t.stride(2)};
+}
+
 }  // namespace
+
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    bool causal, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
+              "attn_fwd: bf16 CUDA tensors required");
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(k.size(2) == S, "cross-attention S_kv != S_q not supported yet");
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
+  auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid((S + QT - 1) / QT, B * H), block(256);
+  Strides qs = get_strides(q), ks = get_strides(k), vs = get_strides(v),
+          os = get_strides(o);
+  const unsigned short* qp = (const unsigned short*)q.data_ptr();
+  const unsigned short* kp = (const unsigned short*)k.data_ptr();
+  const unsigned short* vp = (const unsigned short*)v.data_ptr();
+  unsigned short* op = (unsigned short*)o.data_ptr();
+  float* lp = lse.data_ptr<float>();
+#define LAUNCH(DD, CC)                                                        \
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, block, 0, stream, qp,   \
+                     kp, vp, op, lp, qs, ks, vs, os, B, H, S, (float)scale)
+  if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
+  else          { if (causal) LAUNCH(64, true);  else LAUNCH(64, false);  }
+#undef LAUNCH
+  HIP_CHECK_LAST();
+  return {o, lse};
+}
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
+                                    torch::Tensor dq, torch::Tensor dk,
+                                    torch::Tensor dv,
                                     bool causal, double scale) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(D == 64 || D == 128);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
   auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
+  Strides qs = get_strides(q), ks = get_strides(k), vs = get_strides(v),
+          os = get_strides(o), dos = get_strides(dout),
+          dqs = get_strides(dq), dks = get_strides(dk), dvs = get_strides(dv);
 
-  {  // delta = rowsum(do * o)
+  {
     long rows = (long)B * H * S;
-    int waves_per_block = 4;
-    long blocks = (rows + waves_per_block - 1) / waves_per_block;
+    int wpb = 4;
+    long blocks = (rows + wpb - 1) / wpb;
     hipLaunchKernelGGL(attn_delta_kernel, dim3((unsigned)blocks), dim3(256),
                        0, stream, (const unsigned short*)o.data_ptr(),
-                       (const unsigned short*)dout.contiguous().data_ptr(),
-                       delta.data_ptr<float>(), rows, D);
+                       (const unsigned short*)dout.data_ptr(),
+                       delta.data_ptr<float>(), os, dos, B, H, S, D);
   }
 
   dim3 grid((S + QT - 1) / QT, B * H), block(256);
   const unsigned short* qp = (const unsigned short*)q.data_ptr();
   const unsigned short* kp = (const unsigned short*)k.data_ptr();
   const unsigned short* vp = (const unsigned short*)v.data_ptr();
-  const unsigned short* dop = (const unsigned short*)dout.contiguous().data_ptr();
+  const unsigned short* dop = (const unsigned short*)dout.data_ptr();
   const float* lp = lse.data_ptr<float>();
   const float* delp = delta.data_ptr<float>();
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
     hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, block, 0, stream,  \
                        qp, kp, vp, dop, lp, delp,                             \
-                       (unsigned short*)dq.data_ptr(), B, H, S,               \
-                       (float)scale);                                         \
+                       (unsigned short*)dq.data_ptr(), qs, ks, vs, dos, dqs,  \
+                       B, H, S, (float)scale);                                \
     hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DD, CC>), grid, block, 0,        \
                        stream, qp, kp, vp, dop, lp, delp,                     \
                        (unsigned short*)dk.data_ptr(),                        \
-                       (unsigned short*)dv.data_ptr(), B, H, S,               \
-                       (float)scale);                                         \
+                       (unsigned short*)dv.data_ptr(), qs, ks, vs, dos,       \
+                       dks, dvs, B, H, S, (float)scale);                      \
   } while (0)
   if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
   else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false);  }

@@ -20,11 +20,13 @@ void ema_update(torch::Tensor ema, torch::Tensor p, double decay);
 torch::Tensor l2norm_sq(torch::Tensor x);
 void scale_inplace(torch::Tensor x, double s);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                    torch::Tensor v, bool causal,
-                                    double scale);
+                                    torch::Tensor v, torch::Tensor o,
+                                    bool causal, double scale);
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
+                                    torch::Tensor dq, torch::Tensor dk,
+                                    torch::Tensor dv,
                                     bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 

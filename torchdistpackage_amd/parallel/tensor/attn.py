@@ -18,24 +18,19 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ...ops import flash_attention
+from ...ops import flash_attention, fused_qkv_attention
 from .tp_utils import (ColParallelLinear, RowParallelLinear, TpLinear,
                        copy_to_tp_region, gather_from_sequence_parallel_region,
                        get_tp_size, is_sequence_parallel)
 
 
 def _sdpa(x_qkv: torch.Tensor, n_head: int, causal: bool) -> torch.Tensor:
-    """(S, B, 3*Hl*hd) fused qkv -> (S, B, Hl*hd) attention output."""
-    S, B, three_d = x_qkv.shape
-    d_local = three_d // 3
-    hd = d_local // n_head
-    q, k, v = x_qkv.split(d_local, dim=-1)
+    """(S, B, 3*Hl*hd) fused qkv -> (S, B, Hl*hd) attention output.
 
-    def to_bhsd(t):
-        return t.reshape(S, B, n_head, hd).permute(1, 2, 0, 3).contiguous()
-
-    o = flash_attention(to_bhsd(q), to_bhsd(k), to_bhsd(v), causal=causal)
-    return o.permute(2, 0, 1, 3).reshape(S, B, d_local)
+    Zero-copy path: the flash kernels read/write strided views into the fused
+    qkv / output buffers (no permute-contiguous transposes).
+    """
+    return fused_qkv_attention(x_qkv.contiguous(), n_head, causal=causal)
 
 
 class Attention(nn.Module):
