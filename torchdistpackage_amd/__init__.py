@@ -18,7 +18,8 @@ from .dist import (setup_distributed, tpc, torch_parallel_context,
                    get_mp_ckpt_suffix, mp_ckpt_name, ProcessTopology,
                    hip_prof_start, hip_prof_stop, cu_prof_start, cu_prof_stop,
                    roctx_decorator, nvtx_decorator, ROCTXContext, NVTXContext,
-                   has_inf_or_nan, disable_non_master_print, bench_collectives)
+                   has_inf_or_nan, disable_non_master_print, bench_collectives,
+                   save_checkpoint, load_checkpoint, latest_step)
 from .ddp import (NaiveDdp, NaiveDDP, MoEDP, create_moe_dp_hooks,
                   moe_dp_iter_step, Bf16ZeroOptimizer)
 from .utils import fix_rand, partition_by_numel

@@ -8,3 +8,4 @@ from .utils import (hip_prof_start, hip_prof_stop, cu_prof_start, cu_prof_stop,
                     roctx_decorator, nvtx_decorator, ROCTXContext, NVTXContext,
                     has_inf_or_nan, disable_non_master_print, restore_print)
 from .comm_bench import bench_collectives
+from .checkpoint import save_checkpoint, load_checkpoint, latest_step

@@ -105,6 +105,7 @@ class FusedAdamW(torch.optim.Optimizer):
                 for g in self.param_groups],
         }
 
+    @torch.no_grad()
     def load_state_dict(self, sd):
         if not self._built:
             self._build()

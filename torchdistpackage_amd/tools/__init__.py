@@ -14,3 +14,4 @@ try:
     from .module_replace import replace_linear_by_bminf  # noqa: F401
 except ImportError:
     pass
+from .slurm_monitor import monitor_job, submit_job, job_state
