@@ -205,3 +205,48 @@ def test_l2norm_and_scale():
     xf = x.float()
     ops.scale_(x, 0.5)
     assert torch.allclose(x.float(), xf * 0.5, rtol=1e-2, atol=1e-3)
+
+
+def test_fused_adamw_multi_tensor():
+    """FusedAdamW optimizer (multi-tensor chunked kernel, bf16 params+grads)
+    vs the same math computed in fp32 on CPU."""
+    import torch.nn as nn
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+
+    torch.manual_seed(0)
+    model = nn.Sequential(nn.Linear(67, 129), nn.Linear(129, 31)) \
+        .to(_dev()).to(torch.bfloat16)
+    opt = FusedAdamW(model.parameters(), lr=1e-2, betas=(0.9, 0.95),
+                     eps=1e-8, weight_decay=0.1)
+
+    # CPU fp32 mirror of params
+    cpu_master = [p.detach().float().cpu().clone() for p in model.parameters()]
+    cpu_m = [torch.zeros_like(t) for t in cpu_master]
+    cpu_v = [torch.zeros_like(t) for t in cpu_master]
+
+    for step in range(1, 4):
+        grads = []
+        for p in model.parameters():
+            g = torch.randn_like(p)
+            p.grad = g
+            grads.append(g.float().cpu())
+        opt.step()
+        bc1, bc2 = 1 - 0.9 ** step, 1 - 0.95 ** step
+        for t, m, v, g in zip(cpu_master, cpu_m, cpu_v, grads):
+            t.mul_(1 - 1e-2 * 0.1)
+            m.mul_(0.9).add_(g, alpha=0.1)
+            v.mul_(0.95).addcmul_(g, g, value=0.05)
+            t.addcdiv_(m, (v / bc2).sqrt().add_(1e-8), value=-1e-2 / bc1)
+        opt.zero_grad()
+
+    fg = opt.param_groups[0]["_flat"]
+    assert fg.mt_ready, "multi-tensor path must be active on GPU"
+    off = 0
+    for p, ref in zip(model.parameters(), cpu_master):
+        got = fg.master[off:off + p.numel()].cpu().view_as(ref)
+        assert torch.allclose(got, ref, atol=1e-5), \
+            f"master mismatch {(got - ref).abs().max()}"
+        # bf16 param mirrors master
+        assert torch.allclose(p.detach().float().cpu(),
+                              ref.to(torch.bfloat16).float(), atol=1e-2)
+        off += p.numel()

@@ -32,10 +32,14 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-constexpr int QT = 64;    // q rows (or keys, in dkdv) per workgroup
 constexpr int WQ = 16;    // rows per wave
 constexpr int KT = 32;    // keys (or q rows, in dkdv) per LDS tile
-constexpr int NWAVE = 4;
+// waves per workgroup: the block's Q-tile (or key-tile) is NW*WQ rows; K/V
+// (or Q/dO) tiles are re-read S/(NW*WQ) times, so wider blocks cut the
+// dominant HBM traffic proportionally.
+constexpr int NW_FWD = 8;   // 128-row q-tile
+constexpr int NW_DQ = 8;
+constexpr int NW_DKDV = 8;  // 128-key tile
 
 struct Strides {            // element strides of a (B,H,S,D) view
   long b, h, s;
@@ -58,8 +62,8 @@ DEVINL int swz32(int row, int col) {
 
 // ---------------------------------------------------------------- forward
 
-template <int D, bool CAUSAL>
-__launch_bounds__(256)
+template <int D, bool CAUSAL, int NW>
+__launch_bounds__(NW * 64)
 __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
                                 const unsigned short* __restrict__ k,
                                 const unsigned short* __restrict__ v,
@@ -71,8 +75,9 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
   constexpr int DC = D / 16;
   __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];        // K[key][d], swzD
   __shared__ __attribute__((aligned(16))) unsigned short vt_lds[D * KT];       // V^T[d][key], swz32
-  __shared__ __attribute__((aligned(16))) unsigned short p_lds[NWAVE * 16 * KT];  // P, swz32 per wave
+  __shared__ __attribute__((aligned(16))) unsigned short p_lds[NW * 16 * KT];  // P, swz32 per wave
 
+  constexpr int QT = NW * WQ;
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
   const int qbase = blockIdx.x * QT;
@@ -110,7 +115,7 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
     __syncthreads();
     {
       const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += 256 * 8) {
+      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
         int key = idx / D;
         int col = idx % D;
         int gkey = kt0 + key;
@@ -247,8 +252,8 @@ __global__ void attn_delta_kernel(const unsigned short* __restrict__ o,
   if (lane == 0) delta[row] = acc;
 }
 
-template <int D, bool CAUSAL>
-__launch_bounds__(256)
+template <int D, bool CAUSAL, int NW>
+__launch_bounds__(NW * 64)
 __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                                    const unsigned short* __restrict__ k,
                                    const unsigned short* __restrict__ v,
@@ -264,8 +269,9 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];     // K[key][d], swzD
   __shared__ __attribute__((aligned(16))) unsigned short kt_lds[D * KT];    // K^T[d][key], swz32
   __shared__ __attribute__((aligned(16))) unsigned short v_lds[KT * D];     // V[key][d], swzD
-  __shared__ __attribute__((aligned(16))) unsigned short ds_lds[NWAVE * 16 * KT];  // dS, swz32
+  __shared__ __attribute__((aligned(16))) unsigned short ds_lds[NW * 16 * KT];  // dS, swz32
 
+  constexpr int QT = NW * WQ;
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
   const int qbase = blockIdx.x * QT;
@@ -312,7 +318,7 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
     __syncthreads();
     {
       const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += 256 * 8) {
+      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
         int key = idx / D;
         int col = idx % D;
         int gkey = kt0 + key;
@@ -387,8 +393,8 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   }
 }
 
-template <int D, bool CAUSAL>
-__launch_bounds__(256)
+template <int D, bool CAUSAL, int NW>
+__launch_bounds__(NW * 64)
 __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
                                      const unsigned short* __restrict__ k,
                                      const unsigned short* __restrict__ v,
@@ -406,10 +412,11 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
   __shared__ __attribute__((aligned(16))) unsigned short qt_lds[D * KT];    // Q^T[d][qrow], swz32
   __shared__ __attribute__((aligned(16))) unsigned short do_lds[KT * D];    // dO[qrow][d], swzD
   __shared__ __attribute__((aligned(16))) unsigned short dot_lds[D * KT];   // dO^T[d][qrow], swz32
-  __shared__ __attribute__((aligned(16))) unsigned short st_lds[NWAVE * 16 * KT];  // P^T / dS^T, swz32
+  __shared__ __attribute__((aligned(16))) unsigned short st_lds[NW * 16 * KT];  // P^T / dS^T, swz32
   __shared__ __attribute__((aligned(16))) float lse_lds[KT];
   __shared__ __attribute__((aligned(16))) float del_lds[KT];
 
+  constexpr int QT = NW * WQ;
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
   const int kbase = blockIdx.x * QT;
@@ -452,7 +459,7 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
     __syncthreads();
     {
       const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += 256 * 8) {
+      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
         int row = idx / D;
         int col = idx % D;
         int grow = qt0 + row;
@@ -575,7 +582,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid((S + QT - 1) / QT, B * H), block(256);
+  constexpr int QT = NW_FWD * WQ;
+  dim3 grid((S + QT - 1) / QT, B * H), block(NW_FWD * 64);
   Strides qs = get_strides(q), ks = get_strides(k), vs = get_strides(v),
           os = get_strides(o);
   const unsigned short* qp = (const unsigned short*)q.data_ptr();
@@ -584,8 +592,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   unsigned short* op = (unsigned short*)o.data_ptr();
   float* lp = lse.data_ptr<float>();
 #define LAUNCH(DD, CC)                                                        \
-  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC>), grid, block, 0, stream, qp,   \
-                     kp, vp, op, lp, qs, ks, vs, os, B, H, S, (float)scale)
+  hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, NW_FWD>), grid, block, 0,       \
+                     stream, qp, kp, vp, op, lp, qs, ks, vs, os, B, H, S,     \
+                     (float)scale)
   if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
   else          { if (causal) LAUNCH(64, true);  else LAUNCH(64, false);  }
 #undef LAUNCH
@@ -618,7 +627,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        delta.data_ptr<float>(), os, dos, B, H, S, D);
   }
 
-  dim3 grid((S + QT - 1) / QT, B * H), block(256);
+  constexpr int QT_DQ = NW_DQ * WQ;
+  constexpr int QT_KV = NW_DKDV * WQ;
+  dim3 grid_dq((S + QT_DQ - 1) / QT_DQ, B * H), block_dq(NW_DQ * 64);
+  dim3 grid_kv((S + QT_KV - 1) / QT_KV, B * H), block_kv(NW_DKDV * 64);
   const unsigned short* qp = (const unsigned short*)q.data_ptr();
   const unsigned short* kp = (const unsigned short*)k.data_ptr();
   const unsigned short* vp = (const unsigned short*)v.data_ptr();
@@ -627,12 +639,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   const float* delp = delta.data_ptr<float>();
 #define LAUNCH_BWD(DD, CC)                                                    \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC>), grid, block, 0, stream,  \
-                       qp, kp, vp, dop, lp, delp,                             \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC, NW_DQ>), grid_dq,         \
+                       block_dq, 0, stream, qp, kp, vp, dop, lp, delp,        \
                        (unsigned short*)dq.data_ptr(), qs, ks, vs, dos, dqs,  \
                        B, H, S, (float)scale);                                \
-    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DD, CC>), grid, block, 0,        \
-                       stream, qp, kp, vp, dop, lp, delp,                     \
+    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DD, CC, NW_DKDV>), grid_kv,      \
+                       block_kv, 0, stream, qp, kp, vp, dop, lp, delp,        \
                        (unsigned short*)dk.data_ptr(),                        \
                        (unsigned short*)dv.data_ptr(), qs, ks, vs, dos,       \
                        dks, dvs, B, H, S, (float)scale);                      \

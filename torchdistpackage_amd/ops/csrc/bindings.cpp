@@ -16,6 +16,12 @@ torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
 void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, long step, double lr, double beta1,
                 double beta2, double eps, double wd);
+void multi_adamw_step(torch::Tensor cpid, torch::Tensor coff,
+                      torch::Tensor pptrs, torch::Tensor gptrs,
+                      torch::Tensor moffs, torch::Tensor numels,
+                      torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                      long step, double lr, double beta1, double beta2,
+                      double eps, double wd, bool param_bf16, bool grad_bf16);
 void ema_update(torch::Tensor ema, torch::Tensor p, double decay);
 torch::Tensor l2norm_sq(torch::Tensor x);
 void scale_inplace(torch::Tensor x, double s);
@@ -38,6 +44,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd);
   m.def("bias_gelu_bwd", &bias_gelu_bwd);
   m.def("adamw_step", &adamw_step);
+  m.def("multi_adamw_step", &multi_adamw_step);
   m.def("ema_update", &ema_update);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_inplace", &scale_inplace);

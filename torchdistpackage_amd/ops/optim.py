@@ -23,6 +23,8 @@ from . import fused_adamw_
 
 
 class _FlatGroup:
+    CHUNK = 1 << 16  # must match MT_CHUNK in elementwise.hip
+
     def __init__(self, params: List[torch.Tensor]):
         self.params = params
         n = sum(p.numel() for p in params)
@@ -33,16 +35,36 @@ class _FlatGroup:
         self.exp_avg_sq = torch.zeros(n, dtype=torch.float32, device=dev)
         self.master_views = []
         self.grad_views = []
+        offs = []
         off = 0
         for p in params:
             k = p.numel()
+            offs.append(off)
             mv = self.master.narrow(0, off, k).view_as(p)
             mv.copy_(p.detach().to(torch.float32))
             self.master_views.append(mv)
             self.grad_views.append(self.grad32.narrow(0, off, k).view_as(p))
             off += k
-        self.all_fp32_inplace = all(
-            p.dtype == torch.float32 for p in params)
+        dtypes = {p.dtype for p in params}
+        self.uniform_dtype = params[0].dtype if len(dtypes) == 1 else None
+        # multi-tensor chunk table (GPU, uniform dtype only): one kernel
+        # launch per step, bf16 grads read + bf16 params written in-pass
+        self.mt_ready = False
+        if dev.type == "cuda" and self.uniform_dtype in (
+                torch.bfloat16, torch.float32):
+            cpid, coff = [], []
+            for i, p in enumerate(params):
+                for c0 in range(0, p.numel(), self.CHUNK):
+                    cpid.append(i)
+                    coff.append(c0)
+            self.cpid = torch.tensor(cpid, dtype=torch.int32, device=dev)
+            self.coff = torch.tensor(coff, dtype=torch.int64, device=dev)
+            self.pptrs = torch.tensor([p.data_ptr() for p in params],
+                                      dtype=torch.int64, device=dev)
+            self.moffs = torch.tensor(offs, dtype=torch.int64, device=dev)
+            self.numels = torch.tensor([p.numel() for p in params],
+                                       dtype=torch.int64, device=dev)
+            self.mt_ready = True
 
 
 class FusedAdamW(torch.optim.Optimizer):
@@ -76,13 +98,28 @@ class FusedAdamW(torch.optim.Optimizer):
                 continue
             lr = group["lr"]
             beta1, beta2 = group["betas"]
-            grads = [p.grad if p.grad is not None
-                     else torch.zeros_like(p) for p in fg.params]
-            torch._foreach_copy_(fg.grad_views, grads)
-            fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
-                         self._step, lr, beta1, beta2, group["eps"],
-                         group["weight_decay"])
-            torch._foreach_copy_(fg.params, fg.master_views)
+            if fg.mt_ready:
+                from . import ext
+                gptrs = torch.tensor(
+                    [p.grad.data_ptr() if p.grad is not None else 0
+                     for p in fg.params], dtype=torch.int64)
+                gptrs = gptrs.to(fg.master.device, non_blocking=True)
+                grad_dtype = next((p.grad.dtype for p in fg.params
+                                   if p.grad is not None), fg.uniform_dtype)
+                ext("multi_adamw").multi_adamw_step(
+                    fg.cpid, fg.coff, fg.pptrs, gptrs, fg.moffs, fg.numels,
+                    fg.master, fg.exp_avg, fg.exp_avg_sq, self._step, lr,
+                    beta1, beta2, group["eps"], group["weight_decay"],
+                    fg.uniform_dtype == torch.bfloat16,
+                    grad_dtype == torch.bfloat16)
+            else:
+                grads = [p.grad if p.grad is not None
+                         else torch.zeros_like(p) for p in fg.params]
+                torch._foreach_copy_(fg.grad_views, grads)
+                fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
+                             self._step, lr, beta1, beta2, group["eps"],
+                             group["weight_decay"])
+                torch._foreach_copy_(fg.params, fg.master_views)
         return loss
 
     def zero_grad(self, set_to_none: bool = True):
