@@ -53,9 +53,9 @@ class PPStage(nn.Module):
             mb_size = self.labels_full.shape[0] // self.num_mb
             labels = self.labels_full.narrow(0, self._mb * mb_size, mb_size)
             self._mb += 1
-            loss = F.cross_entropy(
-                logits.reshape(-1, logits.size(-1)).float(),
-                labels.reshape(-1)) / self.num_mb
+            from torchdistpackage_amd.ops import cross_entropy_loss
+            loss = cross_entropy_loss(logits.transpose(0, 1),
+                                      labels.transpose(0, 1)) / self.num_mb
             return loss
         return x
 

@@ -250,3 +250,21 @@ def test_fused_adamw_multi_tensor():
         assert torch.allclose(p.detach().float().cpu(),
                               ref.to(torch.bfloat16).float(), atol=1e-2)
         off += p.numel()
+
+
+def test_fused_cross_entropy():
+    from torchdistpackage_amd.ops import cross_entropy_loss
+    torch.manual_seed(9)
+    N, V = 512, 50304
+    logits = (torch.randn(N, V, device=_dev()) * 3).bfloat16().requires_grad_(True)
+    targets = torch.randint(0, V, (N,), device=_dev())
+    loss = cross_entropy_loss(logits, targets)
+
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, targets)
+    assert abs(loss.item() - ref.item()) < 2e-3, \
+        f"{loss.item()} vs {ref.item()}"
+    loss.backward()
+    ref.backward()
+    err = (logits.grad.float() - lf.grad).abs().max().item()
+    assert err < 1e-4, f"CE dlogits err {err}"

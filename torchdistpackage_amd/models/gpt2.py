@@ -116,9 +116,11 @@ class GPT2Model(nn.Module):
         logits = self.head(x)
         out = {"logits": logits}
         if labels is not None:
-            out["loss"] = F.cross_entropy(
-                logits.reshape(-1, logits.size(-1)).float(),
-                labels.reshape(-1))
+            from ..ops import cross_entropy_loss
+            # logits is a (B,S,V) transpose VIEW of the contiguous (S,B,V)
+            # head output; transposing back avoids a reshape copy in the loss
+            out["loss"] = cross_entropy_loss(logits.transpose(0, 1),
+                                             labels.transpose(0, 1))
         return out
 
     # -- pipeline partitioning support ----------------------------------

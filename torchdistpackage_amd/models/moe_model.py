@@ -81,9 +81,9 @@ class MoEModel(nn.Module):
         logits = self.head(x)
         out = {"logits": logits, "aux_loss": aux_total}
         if labels is not None:
-            ce = F.cross_entropy(
-                logits.reshape(-1, logits.size(-1)).float(),
-                labels.reshape(-1))
+            from ..ops import cross_entropy_loss
+            ce = cross_entropy_loss(logits.transpose(0, 1),
+                                    labels.transpose(0, 1))
             out["loss"] = ce + self.cfg.aux_loss_weight * aux_total.to(ce.dtype)
         return out
 

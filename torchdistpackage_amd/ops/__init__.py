@@ -424,3 +424,32 @@ class _FusedQKVAttentionFn(torch.autograd.Function):
 def fused_qkv_attention(qkv, n_head: int, causal: bool = True,
                         scale: float = None):
     return _FusedQKVAttentionFn.apply(qkv, n_head, causal, scale)
+
+
+class _CrossEntropyFn(torch.autograd.Function):
+    """Fused mean cross-entropy on bf16 logits (one online-LSE pass fwd, one
+    dlogits pass bwd) — replaces logits.float()+F.cross_entropy which
+    materializes fp32 logits.  CPU fallback is exact F.cross_entropy."""
+
+    @staticmethod
+    def forward(ctx, logits2d, targets):
+        loss_vec, lse = ext("cross_entropy").ce_fwd(logits2d, targets)
+        ctx.save_for_backward(logits2d, targets, lse)
+        return loss_vec.mean()
+
+    @staticmethod
+    def backward(ctx, g):
+        logits2d, targets, lse = ctx.saved_tensors
+        dlogits = ext("cross_entropy").ce_bwd(
+            logits2d, targets, lse, g.reshape(1).float().contiguous())
+        return dlogits, None
+
+
+def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor):
+    """Mean CE over flattened (N, V) logits; fused bf16 kernel on GPU."""
+    l2 = logits.reshape(-1, logits.size(-1))
+    t = targets.reshape(-1)
+    if l2.is_cuda and l2.dtype == torch.bfloat16:
+        return _CrossEntropyFn.apply(l2.contiguous(), t)
+    import torch.nn.functional as F
+    return F.cross_entropy(l2.float(), t)

@@ -21,6 +21,7 @@ SOURCES = [
     "norms.hip",
     "elementwise.hip",
     "attention.hip",
+    "cross_entropy.hip",
     "probe.hip",
 ]
 

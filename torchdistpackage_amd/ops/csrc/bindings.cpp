@@ -35,6 +35,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor dv,
                                     bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
+                     torch::Tensor lse, torch::Tensor grad_out);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -51,4 +54,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
 }
