@@ -33,7 +33,7 @@ def parse_args():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=8,
+    ap.add_argument("--batch", type=int, default=16,
                     help="micro/global batch per DP rank")
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--model", type=str, default="gpt2_1.3b",

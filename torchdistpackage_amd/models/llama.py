@@ -116,10 +116,7 @@ class LlamaAttention(nn.Module):
         q = self.rope(shape(q, self.nh_local))
         k = self.rope(shape(k, self.nkv_local))
         v = shape(v, self.nkv_local)
-        rep = self.nh_local // self.nkv_local
-        if rep > 1:
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
+        # GQA handled natively by the flash kernels (kv head = q head // rep)
         o = flash_attention(q, k, v, causal=self.causal)
         o = o.permute(2, 0, 1, 3).reshape(S, B, self.nh_local * self.hd)
         return self.wo(o)

@@ -226,20 +226,26 @@ def bias_gelu(x, bias=None):
 # ---------------------------------------------------------------------------
 
 class _FlashAttentionFn(torch.autograd.Function):
-    """q,k,v: (B, H, S, D) contiguous.  GPU path = in-tree HIP flash kernel
-    (blockwise online-softmax, spec: reference explore/flash-attn/
-    tile_attn.py:100-212); CPU path = exact math attention in fp32."""
+    """q: (B, H, S, D); k, v: (B, H_kv, S, D) with H % H_kv == 0 (GQA).
+    GPU path = in-tree HIP flash kernels (blockwise online-softmax, spec:
+    reference explore/flash-attn/tile_attn.py:100-212); CPU path = exact
+    math attention in fp32 (the numerics oracle)."""
 
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
+        rep = q.shape[1] // k.shape[1]
         if q.is_cuda:
             o = torch.empty(q.shape, dtype=q.dtype, device=q.device)
             o, lse = ext("flash_attention").attn_fwd(
                 q, k, v, o, bool(causal), float(scale))
         else:
-            qf, kf, vf = q.float(), k.float(), v.float()
+            kf = k.float() if rep == 1 else \
+                k.float().repeat_interleave(rep, 1)
+            vf = v.float() if rep == 1 else \
+                v.float().repeat_interleave(rep, 1)
+            qf = q.float()
             s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
             if causal:
                 S = q.shape[-2]
@@ -257,15 +263,27 @@ class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
+        H, Hkv = q.shape[1], k.shape[1]
+        rep = H // Hkv
         if q.is_cuda:
             dq = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-            dk = torch.empty(k.shape, dtype=k.dtype, device=k.device)
-            dv = torch.empty(v.shape, dtype=v.dtype, device=v.device)
+            # GQA: kernels write per-q-head dk/dv partials; summed below
+            dk = torch.empty(q.shape, dtype=k.dtype, device=k.device)
+            dv = torch.empty(q.shape, dtype=v.dtype, device=v.device)
             dq, dk, dv = ext("flash_attention").attn_bwd(
                 do.contiguous(), q, k, v, o, lse, dq, dk, dv,
                 ctx.causal, ctx.scale)
+            if rep > 1:
+                B, _, S, D = q.shape
+                dk = dk.view(B, Hkv, rep, S, D).sum(2).to(k.dtype)
+                dv = dv.view(B, Hkv, rep, S, D).sum(2).to(v.dtype)
         else:
-            qf, kf, vf, dof = q.float(), k.float(), v.float(), do.float()
+            qf = q.float()
+            kf = k.float() if rep == 1 else \
+                k.float().repeat_interleave(rep, 1)
+            vf = v.float() if rep == 1 else \
+                v.float().repeat_interleave(rep, 1)
+            dof = do.float()
             s = torch.matmul(qf, kf.transpose(-1, -2)) * ctx.scale
             if ctx.causal:
                 S = q.shape[-2]
@@ -279,6 +297,10 @@ class _FlashAttentionFn(torch.autograd.Function):
             ds = p * (dp - d) * ctx.scale
             dq = torch.matmul(ds, kf)
             dk = torch.matmul(ds.transpose(-1, -2), qf)
+            if rep > 1:
+                B, _, S, D = q.shape
+                dk = dk.view(B, Hkv, rep, S, D).sum(2)
+                dv = dv.view(B, Hkv, rep, S, D).sum(2)
             dq, dk, dv = dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
         return dq, dk, dv, None, None
 

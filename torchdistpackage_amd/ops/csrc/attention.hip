@@ -70,7 +70,8 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
                                 unsigned short* __restrict__ o,
                                 float* __restrict__ lse,
                                 Strides qs, Strides ks, Strides vs, Strides os,
-                                int B, int H, int S, float scale) {
+                                int B, int H, int S, float scale,
+                                int q_per_kv) {
   constexpr int KC = D / 32;
   constexpr int DC = D / 16;
   __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];        // K[key][d], swzD
@@ -88,9 +89,10 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
+  const int hkv = hh / q_per_kv;
   const unsigned short* qp = q + bb * qs.b + hh * qs.h;
-  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
-  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
 
   const int qrow0 = qbase + wid * WQ;
   bf16x8_v a_q[KC];
@@ -263,7 +265,8 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                                    unsigned short* __restrict__ dq,
                                    Strides qs, Strides ks, Strides vs,
                                    Strides dos, Strides dqs,
-                                   int B, int H, int S, float scale) {
+                                   int B, int H, int S, float scale,
+                                   int q_per_kv) {
   constexpr int KC = D / 32;
   constexpr int DC = D / 16;
   __shared__ __attribute__((aligned(16))) unsigned short k_lds[KT * D];     // K[key][d], swzD
@@ -282,9 +285,10 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
+  const int hkv = hh / q_per_kv;
   const unsigned short* qp = q + bb * qs.b + hh * qs.h;
-  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
-  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
   const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
   const float* lsep = lse + (long)bh * S;
   const float* delp = delta + (long)bh * S;
@@ -405,7 +409,8 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
                                      unsigned short* __restrict__ dv,
                                      Strides qs, Strides ks, Strides vs,
                                      Strides dos, Strides dks, Strides dvs,
-                                     int B, int H, int S, float scale) {
+                                     int B, int H, int S, float scale,
+                                     int q_per_kv) {
   constexpr int KC = D / 32;
   constexpr int DC = D / 16;
   __shared__ __attribute__((aligned(16))) unsigned short q_lds[KT * D];     // Q[qrow][d], swzD
@@ -427,9 +432,10 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
   const int l15 = lane & 15;
   const int lg = lane >> 4;
 
+  const int hkv = hh / q_per_kv;
   const unsigned short* qp = q + bb * qs.b + hh * qs.h;
-  const unsigned short* kp = k + bb * ks.b + hh * ks.h;
-  const unsigned short* vp = v + bb * vs.b + hh * vs.h;
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
   const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
   const float* lsep = lse + (long)bh * S;
   const float* delp = delta + (long)bh * S;
@@ -578,6 +584,9 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
               "attn_fwd: bf16 CUDA tensors required");
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(H % Hkv == 0, "n_head must be a multiple of n_kv_head");
+  const int q_per_kv = H / Hkv;
   TORCH_CHECK(k.size(2) == S, "cross-attention S_kv != S_q not supported yet");
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
@@ -594,7 +603,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 #define LAUNCH(DD, CC)                                                        \
   hipLaunchKernelGGL((attn_fwd_kernel<DD, CC, NW_FWD>), grid, block, 0,       \
                      stream, qp, kp, vp, op, lp, qs, ks, vs, os, B, H, S,     \
-                     (float)scale)
+                     (float)scale, q_per_kv)
   if (D == 128) { if (causal) LAUNCH(128, true); else LAUNCH(128, false); }
   else          { if (causal) LAUNCH(64, true);  else LAUNCH(64, false);  }
 #undef LAUNCH
@@ -610,6 +619,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     bool causal, double scale) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(H % Hkv == 0);
+  const int q_per_kv = H / Hkv;
+  // GQA: dk/dv buffers must be per-Q-HEAD partials (B,H,S,D); the wrapper
+  // sums groups of q_per_kv afterwards
+  TORCH_CHECK(dk.size(1) == H && dv.size(1) == H,
+              "dk/dv must be expanded to H q-heads for GQA");
   TORCH_CHECK(D == 64 || D == 128);
   auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
@@ -642,12 +658,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     hipLaunchKernelGGL((attn_bwd_dq_kernel<DD, CC, NW_DQ>), grid_dq,         \
                        block_dq, 0, stream, qp, kp, vp, dop, lp, delp,        \
                        (unsigned short*)dq.data_ptr(), qs, ks, vs, dos, dqs,  \
-                       B, H, S, (float)scale);                                \
+                       B, H, S, (float)scale, q_per_kv);                      \
     hipLaunchKernelGGL((attn_bwd_dkdv_kernel<DD, CC, NW_DKDV>), grid_kv,      \
                        block_kv, 0, stream, qp, kp, vp, dop, lp, delp,        \
                        (unsigned short*)dk.data_ptr(),                        \
                        (unsigned short*)dv.data_ptr(), qs, ks, vs, dos,       \
-                       dks, dvs, B, H, S, (float)scale);                      \
+                       dks, dvs, B, H, S, (float)scale, q_per_kv);            \
   } while (0)
   if (D == 128) { if (causal) LAUNCH_BWD(128, true); else LAUNCH_BWD(128, false); }
   else          { if (causal) LAUNCH_BWD(64, true);  else LAUNCH_BWD(64, false);  }

@@ -410,6 +410,152 @@ __global__ void layernorm_bwd_bf16v8(const unsigned short* __restrict__ dy,
   }
 }
 
+template <int NC>
+__global__ void rmsnorm_bwd_bf16_reg(const unsigned short* __restrict__ dy,
+                                     const unsigned short* __restrict__ x,
+                                     const unsigned short* __restrict__ w,
+                                     const float* __restrict__ rstd,
+                                     unsigned short* __restrict__ dx,
+                                     float* __restrict__ dw_partial,
+                                     int rows, int D) {
+  __shared__ float scratch[BLOCK / WAVE];
+  float acc_dw[NC][8];
+#pragma unroll
+  for (int c = 0; c < NC; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc_dw[c][j] = 0.f;
+
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + (long)row * D;
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* dxr = dx + (long)row * D;
+    float r = rstd[row];
+    float dot = 0.f;
+    f8 xv[NC], dyv[NC], wv[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      int i = threadIdx.x * 8 + c * BLOCK * 8;
+      if (i < D) {
+        xv[c] = load8f(xr + i);
+        dyv[c] = load8f(dyr + i);
+        wv[c] = load8f(w + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          dot += wv[c].v[j] * dyv[c].v[j] * xv[c].v[j] * r;
+      }
+    }
+    dot = block_sum<BLOCK>(dot, scratch) / D;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      int i = threadIdx.x * 8 + c * BLOCK * 8;
+      if (i < D) {
+        f8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = xv[c].v[j] * r;
+          out.v[j] = r * (wv[c].v[j] * dyv[c].v[j] - xhat * dot);
+          acc_dw[c][j] += dyv[c].v[j] * xhat;
+        }
+        store8f(dxr + i, out);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int c = 0; c < NC; ++c) {
+    int i = threadIdx.x * 8 + c * BLOCK * 8;
+    if (i < D)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&dw_partial[i + j], acc_dw[c][j]);
+  }
+}
+
+template <int NC>
+__global__ void layernorm_bwd_bf16_reg(const unsigned short* __restrict__ dy,
+                                       const unsigned short* __restrict__ x,
+                                       const unsigned short* __restrict__ w,
+                                       const float* __restrict__ mean,
+                                       const float* __restrict__ rstd,
+                                       unsigned short* __restrict__ dx,
+                                       float* __restrict__ dw_partial,
+                                       float* __restrict__ db_partial,
+                                       int rows, int D) {
+  __shared__ float scratch[2 * BLOCK / WAVE];
+  float acc_dw[NC][8], acc_db[NC][8];
+#pragma unroll
+  for (int c = 0; c < NC; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { acc_dw[c][j] = 0.f; acc_db[c][j] = 0.f; }
+
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + (long)row * D;
+    const unsigned short* xr = x + (long)row * D;
+    unsigned short* dxr = dx + (long)row * D;
+    float mu = mean[row], r = rstd[row];
+    float m1 = 0.f, m2 = 0.f;
+    f8 xv[NC], dyv[NC], wv[NC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      int i = threadIdx.x * 8 + c * BLOCK * 8;
+      if (i < D) {
+        xv[c] = load8f(xr + i);
+        dyv[c] = load8f(dyr + i);
+        wv[c] = load8f(w + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = (xv[c].v[j] - mu) * r;
+          float wdy = wv[c].v[j] * dyv[c].v[j];
+          m1 += wdy;
+          m2 += wdy * xhat;
+        }
+      }
+    }
+    {
+      const int wid = threadIdx.x / WAVE;
+      float a = wave_sum(m1), cgr = wave_sum(m2);
+      if ((threadIdx.x & (WAVE - 1)) == 0) {
+        scratch[wid] = a;
+        scratch[BLOCK / WAVE + wid] = cgr;
+      }
+      __syncthreads();
+      m1 = 0.f; m2 = 0.f;
+#pragma unroll
+      for (int i = 0; i < BLOCK / WAVE; ++i) {
+        m1 += scratch[i];
+        m2 += scratch[BLOCK / WAVE + i];
+      }
+    }
+    m1 /= D;
+    m2 /= D;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      int i = threadIdx.x * 8 + c * BLOCK * 8;
+      if (i < D) {
+        f8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = (xv[c].v[j] - mu) * r;
+          out.v[j] = r * (wv[c].v[j] * dyv[c].v[j] - m1 - xhat * m2);
+          acc_dw[c][j] += dyv[c].v[j] * xhat;
+          acc_db[c][j] += dyv[c].v[j];
+        }
+        store8f(dxr + i, out);
+      }
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int c = 0; c < NC; ++c) {
+    int i = threadIdx.x * 8 + c * BLOCK * 8;
+    if (i < D)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(&dw_partial[i + j], acc_dw[c][j]);
+        atomicAdd(&db_partial[i + j], acc_db[c][j]);
+      }
+  }
+}
+
 }  // namespace
 
 // ------------------------------------------------------------------ C++ API
@@ -455,7 +601,21 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
   size_t lds = (D + BLOCK / WAVE) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0 &&
+      D <= 4 * BLOCK * 8) {
+    int nc = (D + BLOCK * 8 - 1) / (BLOCK * 8);
+#define RMS_REG(NC)                                                          \
+    hipLaunchKernelGGL(rmsnorm_bwd_bf16_reg<NC>, grid, block, 0, stream,     \
+                       (const unsigned short*)dy.data_ptr(),                 \
+                       (const unsigned short*)x.data_ptr(),                  \
+                       (const unsigned short*)w.data_ptr(),                  \
+                       rstd.data_ptr<float>(),                               \
+                       (unsigned short*)dx.data_ptr(),                       \
+                       dw_partial.data_ptr<float>(), (int)rows, D)
+    if (nc == 1) RMS_REG(1); else if (nc == 2) RMS_REG(2);
+    else if (nc == 3) RMS_REG(3); else RMS_REG(4);
+#undef RMS_REG
+  } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(rmsnorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
                        (const unsigned short*)x.data_ptr(),
@@ -527,7 +687,22 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
   size_t lds = (2 * D + 2 * (BLOCK / WAVE)) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
+  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0 &&
+      D <= 4 * BLOCK * 8) {
+    int nc = (D + BLOCK * 8 - 1) / (BLOCK * 8);
+#define LN_REG(NC)                                                           \
+    hipLaunchKernelGGL(layernorm_bwd_bf16_reg<NC>, grid, block, 0, stream,   \
+                       (const unsigned short*)dy.data_ptr(),                 \
+                       (const unsigned short*)x.data_ptr(),                  \
+                       (const unsigned short*)w.data_ptr(),                  \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                       (unsigned short*)dx.data_ptr(),                       \
+                       dw_partial.data_ptr<float>(),                         \
+                       db_partial.data_ptr<float>(), (int)rows, D)
+    if (nc == 1) LN_REG(1); else if (nc == 2) LN_REG(2);
+    else if (nc == 3) LN_REG(3); else LN_REG(4);
+#undef LN_REG
+  } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(layernorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
                        (const unsigned short*)x.data_ptr(),
