@@ -52,7 +52,7 @@ def llama3_8b() -> LlamaConfig:
 
 def llama_tiny() -> LlamaConfig:
     return LlamaConfig(vocab_size=512, n_layer=2, n_head=4, n_kv_head=2,
-                       dim=128, ffn_dim=256, max_seq=128)
+                       dim=256, ffn_dim=512, max_seq=128)
 
 
 class Rope(nn.Module):
@@ -63,7 +63,7 @@ class Rope(nn.Module):
         super().__init__()
         inv = 1.0 / (theta ** (torch.arange(0, head_dim, 2).float() / head_dim))
         t = torch.arange(max_seq).float()
-        freqs = torch.outer(t, inv)                     # (S, hd/2)
+        freqs = torch.outer(t, inv).to(device)          # (S, hd/2)
         self.register_buffer("cos", freqs.cos(), persistent=False)
         self.register_buffer("sin", freqs.sin(), persistent=False)
 

@@ -215,7 +215,7 @@ class ExpertParallelMoE(nn.Module):
         # local expert across src ranks
         outs = torch.empty_like(received)
         # offsets of each (src, local_expert) segment in `received`
-        seg_sizes = my_slice.reshape(-1)          # (ep * num_local,)
+        seg_sizes = my_slice.reshape(-1).cpu()    # (ep * num_local,)
         seg_offs = torch.cumsum(
             torch.cat([torch.zeros(1, dtype=seg_sizes.dtype), seg_sizes]),
             0).tolist()

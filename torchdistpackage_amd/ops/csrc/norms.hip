@@ -461,12 +461,15 @@ __global__ void rmsnorm_bwd_bf16_reg(const unsigned short* __restrict__ dy,
     }
     __syncthreads();
   }
+  // per-block partial row (coalesced plain stores; summed host-side) —
+  // atomicAdd here serializes gridDim-deep per element and was 2x slower
+  float* dwp = dw_partial + (long)blockIdx.x * D;
 #pragma unroll
   for (int c = 0; c < NC; ++c) {
     int i = threadIdx.x * 8 + c * BLOCK * 8;
     if (i < D)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) atomicAdd(&dw_partial[i + j], acc_dw[c][j]);
+      for (int j = 0; j < 8; ++j) dwp[i + j] = acc_dw[c][j];
   }
 }
 
@@ -544,14 +547,16 @@ __global__ void layernorm_bwd_bf16_reg(const unsigned short* __restrict__ dy,
     }
     __syncthreads();
   }
+  float* dwp = dw_partial + (long)blockIdx.x * D;
+  float* dbp = db_partial + (long)blockIdx.x * D;
 #pragma unroll
   for (int c = 0; c < NC; ++c) {
     int i = threadIdx.x * 8 + c * BLOCK * 8;
     if (i < D)
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        atomicAdd(&dw_partial[i + j], acc_dw[c][j]);
-        atomicAdd(&db_partial[i + j], acc_db[c][j]);
+        dwp[i + j] = acc_dw[c][j];
+        dbp[i + j] = acc_db[c][j];
       }
   }
 }
@@ -597,12 +602,15 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int D = x.size(-1);
   const long rows = x.numel() / D;
   auto dx = torch::empty_like(x);
-  auto dw_partial = torch::zeros({D}, x.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
   size_t lds = (D + BLOCK / WAVE) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0 &&
-      D <= 4 * BLOCK * 8) {
+  const bool reg_path = x.scalar_type() == torch::kBFloat16 &&
+      D % 8 == 0 && D <= 4 * BLOCK * 8;
+  auto dw_partial = reg_path
+      ? torch::empty({(long)grid.x, (long)D}, x.options().dtype(torch::kFloat))
+      : torch::zeros({D}, x.options().dtype(torch::kFloat));
+  if (reg_path) {
     int nc = (D + BLOCK * 8 - 1) / (BLOCK * 8);
 #define RMS_REG(NC)                                                          \
     hipLaunchKernelGGL(rmsnorm_bwd_bf16_reg<NC>, grid, block, 0, stream,     \
@@ -615,6 +623,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
     if (nc == 1) RMS_REG(1); else if (nc == 2) RMS_REG(2);
     else if (nc == 3) RMS_REG(3); else RMS_REG(4);
 #undef RMS_REG
+    HIP_CHECK_LAST();
+    return {dx, dw_partial.sum(0).to(w.scalar_type())};
   } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(rmsnorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
@@ -682,13 +692,18 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int D = x.size(-1);
   const long rows = x.numel() / D;
   auto dx = torch::empty_like(x);
-  auto dw_partial = torch::zeros({D}, x.options().dtype(torch::kFloat));
-  auto db_partial = torch::zeros({D}, x.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(pick_grid(rows)), block(BLOCK);
   size_t lds = (2 * D + 2 * (BLOCK / WAVE)) * sizeof(float);
-  if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0 &&
-      D <= 4 * BLOCK * 8) {
+  const bool reg_path = x.scalar_type() == torch::kBFloat16 &&
+      D % 8 == 0 && D <= 4 * BLOCK * 8;
+  auto dw_partial = reg_path
+      ? torch::empty({(long)grid.x, (long)D}, x.options().dtype(torch::kFloat))
+      : torch::zeros({D}, x.options().dtype(torch::kFloat));
+  auto db_partial = reg_path
+      ? torch::empty({(long)grid.x, (long)D}, x.options().dtype(torch::kFloat))
+      : torch::zeros({D}, x.options().dtype(torch::kFloat));
+  if (reg_path) {
     int nc = (D + BLOCK * 8 - 1) / (BLOCK * 8);
 #define LN_REG(NC)                                                           \
     hipLaunchKernelGGL(layernorm_bwd_bf16_reg<NC>, grid, block, 0, stream,   \
@@ -702,6 +717,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
     if (nc == 1) LN_REG(1); else if (nc == 2) LN_REG(2);
     else if (nc == 3) LN_REG(3); else LN_REG(4);
 #undef LN_REG
+    HIP_CHECK_LAST();
+    return {dx, dw_partial.sum(0).to(w.scalar_type()),
+            db_partial.sum(0).to(w.scalar_type())};
   } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(layernorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
