@@ -120,6 +120,66 @@ def partition_balanced(layers: Sequence[nn.Module],
     return best
 
 
+def partition_by_time(layers: Sequence[nn.Module], num_stages: int,
+                      sample_input, warmup: int = 2,
+                      iters: int = 5) -> List[List[int]]:
+    """Contiguous partition balanced by MEASURED per-layer forward time.
+
+    Capability parity with the reference's fx-based auto-split
+    (/root/reference/explore/fx/fx_graph_split.py:123-172), without fx: the
+    layer list is executed sequentially on ``sample_input`` and timed (device
+    sync per layer), then the same bottleneck binary search as
+    partition_balanced slices by time.
+    """
+    import time as _time
+    import torch as _torch
+    x = sample_input
+    times = [0.0] * len(layers)
+    with _torch.no_grad():
+        for it in range(warmup + iters):
+            xi = x
+            for li, layer in enumerate(layers):
+                if _torch.cuda.is_available():
+                    _torch.cuda.synchronize()
+                t0 = _time.perf_counter()
+                xi = layer(xi)
+                if _torch.cuda.is_available():
+                    _torch.cuda.synchronize()
+                if it >= warmup:
+                    times[li] += _time.perf_counter() - t0
+    # reuse the bottleneck search with time weights (scaled to ints)
+    scale = 1e7
+    weights = [max(int(t * scale), 1) for t in times]
+    n = len(weights)
+
+    def feasible(cap):
+        parts, start, acc = [], 0, 0
+        for i, w in enumerate(weights):
+            if w > cap:
+                return None
+            if acc + w > cap:
+                parts.append([start, i])
+                start, acc = i, 0
+            acc += w
+        parts.append([start, n])
+        if len(parts) > num_stages:
+            return None
+        while len(parts) < num_stages:
+            parts.append([n, n])
+        return parts
+
+    lo, hi = max(weights), sum(weights)
+    best = feasible(hi)
+    while lo <= hi:
+        mid = (lo + hi) // 2
+        f = feasible(mid)
+        if f is not None:
+            best, hi = f, mid - 1
+        else:
+            lo = mid + 1
+    return best
+
+
 def flat_and_partition(model: nn.Module, num_stages: Optional[int] = None,
                        method: str = "uniform",
                        exec_order: Optional[Sequence] = None) -> nn.Sequential:
