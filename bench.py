@@ -150,11 +150,17 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
     x = torch.randint(0, cfg.vocab_size, (args.batch, args.seq),
                       generator=g).to(dev)
 
+    from torchdistpackage_amd.parallel.tensor import \
+        allreduce_sequence_parallel_grads
+
     def step():
         out = model(x, labels=x)
         out["loss"].backward()
         if isinstance(model, NaiveDdp):
             model.reduce_gradients()
+        if tp > 1:
+            allreduce_sequence_parallel_grads(
+                model.module if isinstance(model, NaiveDdp) else model)
         opt.step()
         opt.zero_grad()
 

@@ -165,3 +165,54 @@ def test_moe_model_forward():
     assert n_expert == 2 * 4 * 4  # layers * experts * (2 linears w+b)
     for p in m.expert_parameters():
         assert p.grad is not None
+
+
+def _moe_full_composition(rank, world_size):
+    """dp4 = moe_dp2 x moe_ep2 full training composition: dense params sync
+    over 'data' (NaiveDdp), expert params over 'moe_dp' (MoEDP); two steps
+    run and the replicated expert shards stay identical across moe_dp peers."""
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.ddp import NaiveDdp, create_moe_dp_hooks, \
+        moe_dp_iter_step
+    from torchdistpackage_amd.models.moe_model import MoEConfig, MoEModel
+
+    tpc.setup_process_groups([("data", world_size)])
+    tpc.build_moe_groups(moe_dp_size=2, moe_ep_size=2)
+    cfg = MoEConfig(vocab_size=64, n_layer=1, n_head=2, dim=16, max_seq=8,
+                    num_experts=4, top_k=1, hidden_mult=2)
+    torch.manual_seed(0)  # same init; broadcasts enforce it anyway
+    model = MoEModel(cfg)
+    ddp = NaiveDdp(model)
+    create_moe_dp_hooks(list(model.expert_parameters()))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+
+    for it in range(2):
+        torch.manual_seed(it * 10 + rank)
+        x = torch.randint(0, 64, (2, 8))
+        out = ddp(x, labels=x)
+        out["loss"].backward()
+        ddp.reduce_gradients()
+        moe_dp_iter_step()
+        opt.step()
+        opt.zero_grad()
+
+    # expert params must be identical across the moe_dp group
+    grp = tpc.get_group("moe_dp")
+    for p in model.expert_parameters():
+        mine = p.detach().clone()
+        peers = [torch.zeros_like(mine) for _ in range(2)]
+        dist.all_gather(peers, mine, group=grp)
+        assert torch.allclose(peers[0], peers[1], atol=1e-6)
+    # dense params identical across the FULL data group
+    for p in model.non_expert_parameters():
+        mine = p.detach().clone()
+        peers = [torch.zeros_like(mine) for _ in range(world_size)]
+        dist.all_gather(peers, mine)
+        for pr in peers[1:]:
+            assert torch.allclose(peers[0], pr, atol=1e-6)
+    return True
+
+
+def test_moe_full_composition_world4():
+    run_distributed(_moe_full_composition, world_size=4)

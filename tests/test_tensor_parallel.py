@@ -199,3 +199,56 @@ def _tp_gpt2(rank, world_size):
 
 def test_tp_gpt2_model():
     run_distributed(_tp_gpt2, world_size=2)
+
+
+def _tp_llama(rank, world_size):
+    """Llama block at tp=2 (SP on) vs the tp=1 oracle."""
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.tensor import set_tp_group
+    from torchdistpackage_amd.models.llama import (LlamaConfig, LlamaModel)
+
+    cfg = LlamaConfig(vocab_size=128, n_layer=2, n_head=4, n_kv_head=2,
+                      dim=64, ffn_dim=128, max_seq=16)
+    torch.manual_seed(0)
+    full = LlamaModel(cfg)
+    torch.manual_seed(2)
+    x = torch.randint(0, 128, (2, 16))
+    loss_full = full(x, labels=x)["loss"]
+    loss_full.backward()
+
+    tpc.setup_process_groups([("tensor", world_size)])
+    set_tp_group(tpc.get_group("tensor"))
+    torch.manual_seed(0)
+    tp = LlamaModel(cfg)
+    # weight surgery: col shards rows, row shards cols
+    tp.embed.load_state_dict(full.embed.state_dict())
+    tp.head.norm.load_state_dict(full.head.norm.state_dict())
+    with torch.no_grad():
+        tp.head.weight.copy_(full.head.weight)
+        for fb, tb in zip(full.blocks, tp.blocks):
+            tb.attn_norm.load_state_dict(fb.attn_norm.state_dict())
+            tb.mlp_norm.load_state_dict(fb.mlp_norm.state_dict())
+            tb.attn.wq.init_weight_from_full(fb.attn.wq.weight)
+            tb.attn.wk.init_weight_from_full(fb.attn.wk.weight)
+            tb.attn.wv.init_weight_from_full(fb.attn.wv.weight)
+            tb.attn.wo.init_weight_from_full(fb.attn.wo.weight)
+            tb.mlp.w1.init_weight_from_full(fb.mlp.w1.weight)
+            tb.mlp.w3.init_weight_from_full(fb.mlp.w3.weight)
+            tb.mlp.w2.init_weight_from_full(fb.mlp.w2.weight)
+
+    loss_tp = tp(x, labels=x)["loss"]
+    assert torch.allclose(loss_tp, loss_full, atol=1e-4), \
+        f"{float(loss_tp)} vs {float(loss_full)}"
+    loss_tp.backward()
+    from torchdistpackage_amd.parallel.tensor import \
+        allreduce_sequence_parallel_grads
+    allreduce_sequence_parallel_grads(tp)
+    for fb, tb in zip(full.blocks, tp.blocks):
+        assert torch.allclose(tb.attn_norm.weight.grad,
+                              fb.attn_norm.weight.grad, atol=1e-4)
+    return True
+
+
+def test_tp_llama_model():
+    run_distributed(_tp_llama, world_size=2)
