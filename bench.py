@@ -37,7 +37,11 @@ def parse_args():
                     help="micro/global batch per DP rank")
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--model", type=str, default="gpt2_1.3b",
-                    choices=["gpt2_1.3b", "gpt2_small", "tiny"])
+                    choices=["gpt2_1.3b", "gpt2_small", "tiny", "llama_8b",
+                             "moe_8x"])
+    ap.add_argument("--zero", action="store_true",
+                    help="use Bf16ZeroOptimizer (hybrid node-local shard) + "
+                         "sharded EMA instead of plain FusedAdamW")
     ap.add_argument("--micro-batches", type=int, default=8,
                     help="micro-batches per step when PP is active")
     return ap.parse_args()
@@ -85,6 +89,12 @@ def main():
         cfg = gpt2_xl_1p3b()
     elif args.model == "gpt2_small":
         cfg = gpt2_small()
+    elif args.model == "llama_8b":
+        from torchdistpackage_amd.models.llama import llama3_8b
+        cfg = llama3_8b()
+    elif args.model == "moe_8x":
+        from torchdistpackage_amd.models.moe_model import mixtral_style_8x
+        cfg = mixtral_style_8x()
     else:
         cfg = GPT2Config(vocab_size=2048, n_layer=4, n_head=8, dim=512,
                          max_seq=args.seq)
@@ -138,11 +148,41 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
 
     world = dist.get_world_size() if dist.is_initialized() else 1
     torch.manual_seed(1234)  # same init across ranks (then broadcast anyway)
-    model = GPT2Model(cfg, device=dev, dtype=dtype)
+    if args.model == "llama_8b":
+        from torchdistpackage_amd.models.llama import LlamaModel
+        model = LlamaModel(cfg, device=dev, dtype=dtype)
+    elif args.model == "moe_8x":
+        from torchdistpackage_amd.models.moe_model import MoEModel
+        model = MoEModel(cfg, device=dev, dtype=dtype)
+        if world > 1:
+            ep = min(dp, cfg.num_experts)
+            tpc.build_moe_groups(moe_dp_size=dp // ep, moe_ep_size=ep)
+    else:
+        model = GPT2Model(cfg, device=dev, dtype=dtype)
 
+    ema = None
     if world > 1 and dp > 1:
         model = NaiveDdp(model, group=tpc.get_group("data"))
-    opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
+        if args.model == "moe_8x":
+            from torchdistpackage_amd.ddp import create_moe_dp_hooks
+            inner = model.module
+            if tpc.get_group_size("moe_dp") > 1:
+                create_moe_dp_hooks(list(inner.expert_parameters()))
+    if args.zero:
+        from torchdistpackage_amd import Bf16ZeroOptimizer, ShardedEMA, \
+            setup_node_groups
+        inner_model = model.module if isinstance(model, NaiveDdp) else model
+        node_group = setup_node_groups(num_per_node=8) if world > 1 else None
+        inner_opt = FusedAdamW(inner_model.parameters(), lr=1e-4,
+                               weight_decay=0.1)
+        opt = Bf16ZeroOptimizer(inner_opt, group=node_group,
+                                grad_group=tpc.get_group("data")
+                                if world > 1 else None, stage2=True)
+        ema = ShardedEMA(inner_model, decay=0.999)
+        # ZeRO owns grad reduction: unhook NaiveDdp if wrapped
+        model = inner_model
+    else:
+        opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
 
     # identical data inside a TP group; different across DP ranks
     dp_rank = tpc.get_dp_rank() if world > 1 else 0
@@ -158,11 +198,16 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         out["loss"].backward()
         if isinstance(model, NaiveDdp):
             model.reduce_gradients()
+            from torchdistpackage_amd.ddp import moe_dp_iter_step
+            if args.model == "moe_8x":
+                moe_dp_iter_step()
         if tp > 1:
             allreduce_sequence_parallel_grads(
                 model.module if isinstance(model, NaiveDdp) else model)
         opt.step()
         opt.zero_grad()
+        if ema is not None:
+            ema.update()
 
     for _ in range(args.warmup):
         step()
