@@ -37,7 +37,7 @@ constexpr int KT = 32;    // keys (or q rows, in dkdv) per LDS tile
 // waves per workgroup: the block's Q-tile (or key-tile) is NW*WQ rows; K/V
 // (or Q/dO) tiles are re-read S/(NW*WQ) times, so wider blocks cut the
 // dominant HBM traffic proportionally.
-constexpr int NW_FWD = 8;   // 128-row q-tile
+constexpr int NW_FWD = 16;  // 256-row q-tile
 constexpr int NW_DQ = 8;
 constexpr int NW_DKDV = 8;  // 128-key tile
 
@@ -113,32 +113,41 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ q,
 
   const int kv_end = CAUSAL ? min(S, qbase + QT) : S;
 
+  // T14 async-STAGE split: each thread owns ONE 8-elem chunk of the K and V
+  // tiles (KT*D == NW*64*8 at D=128); the next tile's global loads are
+  // issued right after the LDS write barrier, so they fly under the MFMA
+  // compute of the current tile (guide 6.15: +17% on attention).
+  const int st_idx = tid * 8;
+  const int st_key = st_idx / D;
+  const int st_col = st_idx % D;
+  const bool st_own = st_idx < KT * D;
+  bf16x8_v k_reg, v_reg;
+
+  auto stage_load = [&](int kt0) {
+    if (!st_own) return;
+    int gkey = kt0 + st_key;
+    if (gkey < S) {
+      k_reg = pack8(kp + (long)gkey * ks.s + st_col);
+      v_reg = pack8(vp + (long)gkey * vs.s + st_col);
+    } else {
+      k_reg = (bf16x8_v)(__bf16)0.f;
+      v_reg = (bf16x8_v)(__bf16)0.f;
+    }
+  };
+
+  stage_load(0);
+
   for (int kt0 = 0; kt0 < kv_end; kt0 += KT) {
     __syncthreads();
-    {
-      const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
-        int key = idx / D;
-        int col = idx % D;
-        int gkey = kt0 + key;
-        if (gkey < S) {
-          bf16x8_v kv8 = pack8(kp + (long)gkey * ks.s + col);
-          *(bf16x8_v*)&k_lds[swzD<D>(key, col)] = kv8;
-          bf16x8_v vv8 = pack8(vp + (long)gkey * vs.s + col);
-          const unsigned short* vsrc = (const unsigned short*)&vv8;
+    if (st_own) {
+      *(bf16x8_v*)&k_lds[swzD<D>(st_key, st_col)] = k_reg;
+      const unsigned short* vsrc = (const unsigned short*)&v_reg;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vt_lds[swz32(col + j, key)] = vsrc[j];
-        } else {
-          for (int j = 0; j < 8; ++j) {
-            k_lds[swzD<D>(key, col + j - (col + j) % 8) +
-                  ((col + j) % 8)] = 0;  // keep swizzle chunk-aligned
-            vt_lds[swz32(col + j, key)] = 0;
-          }
-        }
-      }
+      for (int j = 0; j < 8; ++j)
+        vt_lds[swz32(st_col + j, st_key)] = vsrc[j];
     }
     __syncthreads();
+    if (kt0 + KT < kv_end) stage_load(kt0 + KT);
 
     f32x4 s_acc[KT / 16];
 #pragma unroll
@@ -318,33 +327,39 @@ __global__ void attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 
   const int kv_end = CAUSAL ? min(S, qbase + QT) : S;
 
+  // T14 async-STAGE split (see fwd kernel)
+  const int st_idx = tid * 8;
+  const int st_key = st_idx / D;
+  const int st_col = st_idx % D;
+  const bool st_own = st_idx < KT * D;
+  bf16x8_v k_reg, v_reg;
+
+  auto stage_load = [&](int kt0) {
+    if (!st_own) return;
+    int gkey = kt0 + st_key;
+    if (gkey < S) {
+      k_reg = pack8(kp + (long)gkey * ks.s + st_col);
+      v_reg = pack8(vp + (long)gkey * vs.s + st_col);
+    } else {
+      k_reg = (bf16x8_v)(__bf16)0.f;
+      v_reg = (bf16x8_v)(__bf16)0.f;
+    }
+  };
+
+  stage_load(0);
+
   for (int kt0 = 0; kt0 < kv_end; kt0 += KT) {
     __syncthreads();
-    {
-      const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
-        int key = idx / D;
-        int col = idx % D;
-        int gkey = kt0 + key;
-        if (gkey < S) {
-          bf16x8_v kv8 = pack8(kp + (long)gkey * ks.s + col);
-          *(bf16x8_v*)&k_lds[swzD<D>(key, col)] = kv8;
-          bf16x8_v vv8 = pack8(vp + (long)gkey * vs.s + col);
-          *(bf16x8_v*)&v_lds[swzD<D>(key, col)] = vv8;
-          const unsigned short* ksrc = (const unsigned short*)&kv8;
+    if (st_own) {
+      *(bf16x8_v*)&k_lds[swzD<D>(st_key, st_col)] = k_reg;
+      *(bf16x8_v*)&v_lds[swzD<D>(st_key, st_col)] = v_reg;
+      const unsigned short* ksrc = (const unsigned short*)&k_reg;
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            kt_lds[swz32(col + j, key)] = ksrc[j];
-        } else {
-          for (int j = 0; j < 8; ++j) {
-            k_lds[swzD<D>(key, col) + j] = 0;
-            v_lds[swzD<D>(key, col) + j] = 0;
-            kt_lds[swz32(col + j, key)] = 0;
-          }
-        }
-      }
+      for (int j = 0; j < 8; ++j)
+        kt_lds[swz32(st_col + j, st_key)] = ksrc[j];
     }
     __syncthreads();
+    if (kt0 + KT < kv_end) stage_load(kt0 + KT);
 
     f32x4 s_acc[KT / 16], dp_acc[KT / 16];
 #pragma unroll
@@ -461,42 +476,53 @@ __global__ void attn_bwd_dkdv_kernel(const unsigned short* __restrict__ q,
 
   const int qt_start = CAUSAL ? kbase : 0;
 
-  for (int qt0 = qt_start; qt0 < S; qt0 += KT) {
-    __syncthreads();
-    {
-      const int elems = KT * D;
-      for (int idx = tid * 8; idx < elems; idx += NW * 64 * 8) {
-        int row = idx / D;
-        int col = idx % D;
-        int grow = qt0 + row;
-        if (grow < S) {
-          bf16x8_v qv8 = pack8(qp + (long)grow * qs.s + col);
-          *(bf16x8_v*)&q_lds[swzD<D>(row, col)] = qv8;
-          bf16x8_v dv8 = pack8(dop + (long)grow * dos.s + col);
-          *(bf16x8_v*)&do_lds[swzD<D>(row, col)] = dv8;
-          const unsigned short* qsrc = (const unsigned short*)&qv8;
-          const unsigned short* dsrc = (const unsigned short*)&dv8;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            qt_lds[swz32(col + j, row)] = qsrc[j];
-            dot_lds[swz32(col + j, row)] = dsrc[j];
-          }
-        } else {
-          for (int j = 0; j < 8; ++j) {
-            q_lds[swzD<D>(row, col) + j] = 0;
-            do_lds[swzD<D>(row, col) + j] = 0;
-            qt_lds[swz32(col + j, row)] = 0;
-            dot_lds[swz32(col + j, row)] = 0;
-          }
-        }
-      }
-      if (tid < KT) {
-        int grow = qt0 + tid;
-        lse_lds[tid] = grow < S ? lsep[grow] : 0.f;
-        del_lds[tid] = grow < S ? delp[grow] : 0.f;
+  // T14 async-STAGE split (see fwd kernel)
+  const int st_idx = tid * 8;
+  const int st_row = st_idx / D;
+  const int st_col = st_idx % D;
+  const bool st_own = st_idx < KT * D;
+  bf16x8_v q_reg, do_reg;
+  float lse_reg = 0.f, del_reg = 0.f;
+
+  auto stage_load = [&](int qt0) {
+    if (st_own) {
+      int grow = qt0 + st_row;
+      if (grow < S) {
+        q_reg = pack8(qp + (long)grow * qs.s + st_col);
+        do_reg = pack8(dop + (long)grow * dos.s + st_col);
+      } else {
+        q_reg = (bf16x8_v)(__bf16)0.f;
+        do_reg = (bf16x8_v)(__bf16)0.f;
       }
     }
+    if (tid < KT) {
+      int grow = qt0 + tid;
+      lse_reg = grow < S ? lsep[grow] : 0.f;
+      del_reg = grow < S ? delp[grow] : 0.f;
+    }
+  };
+
+  stage_load(qt_start);
+
+  for (int qt0 = qt_start; qt0 < S; qt0 += KT) {
     __syncthreads();
+    if (st_own) {
+      *(bf16x8_v*)&q_lds[swzD<D>(st_row, st_col)] = q_reg;
+      *(bf16x8_v*)&do_lds[swzD<D>(st_row, st_col)] = do_reg;
+      const unsigned short* qsrc = (const unsigned short*)&q_reg;
+      const unsigned short* dsrc = (const unsigned short*)&do_reg;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        qt_lds[swz32(st_col + j, st_row)] = qsrc[j];
+        dot_lds[swz32(st_col + j, st_row)] = dsrc[j];
+      }
+    }
+    if (tid < KT) {
+      lse_lds[tid] = lse_reg;
+      del_lds[tid] = del_reg;
+    }
+    __syncthreads();
+    if (qt0 + KT < S) stage_load(qt0 + KT);
 
     f32x4 st_acc[KT / 16], dpt_acc[KT / 16];
 #pragma unroll
