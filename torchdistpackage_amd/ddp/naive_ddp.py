@@ -89,9 +89,15 @@ class NaiveDdp(nn.Module):
 
     def __init__(self, module: nn.Module, group: Optional[dist.ProcessGroup] = None,
                  sync: bool = False, bucket_cap_mb: float = 50.0,
-                 num_grad_acc_iter: int = 1, broadcast_params: bool = True):
+                 num_grad_acc_iter: int = 1, broadcast_params: bool = True,
+                 reduce_op: str = "avg"):
         super().__init__()
         self.module = module
+        # reference parity: reduce_op "avg" (default) or "sum"; the
+        # reference's selector is broken (reduce_op.lower missing parens,
+        # naive_ddp.py:53 — "sum" is never chosen there)
+        assert reduce_op in ("avg", "sum"), reduce_op
+        self.reduce_op = reduce_op
         if group is None:
             try:
                 from ..dist.topo import tpc
@@ -149,6 +155,9 @@ class NaiveDdp(nn.Module):
     def _world(self) -> int:
         return dist.get_world_size(self.group) if dist.is_initialized() else 1
 
+    def _avg(self) -> bool:
+        return self.reduce_op == "avg"
+
     def _build_buckets(self):
         """Pack params into buckets in reverse registration order (grads become
         ready roughly back-to-front during backward)."""
@@ -187,11 +196,12 @@ class NaiveDdp(nn.Module):
         if fires % self.num_grad_acc_iter != 0:
             return  # intermediate micro-batch: accumulate only
         if self.sync:
-            if self._use_gpu:
+            if self._use_gpu and self._avg():
                 dist.all_reduce(p.grad, op=dist.ReduceOp.AVG, group=self.group)
-            else:  # gloo has no AVG
+            else:  # gloo has no AVG; or reduce_op == "sum"
                 dist.all_reduce(p.grad, op=dist.ReduceOp.SUM, group=self.group)
-                p.grad.div_(self._world())
+                if self._avg():
+                    p.grad.div_(self._world())
             return
         bucket, idx = self._param_bucket[id(p)]
         if self._use_gpu:
@@ -211,7 +221,9 @@ class NaiveDdp(nn.Module):
         ev.record(cur)
         with torch.cuda.stream(self._reduce_stream):
             self._reduce_stream.wait_event(ev)
-            dist.all_reduce(bucket.data, op=dist.ReduceOp.AVG, group=self.group)
+            dist.all_reduce(bucket.data,
+                            op=dist.ReduceOp.AVG if self._avg()
+                            else dist.ReduceOp.SUM, group=self.group)
             done = torch.cuda.Event()
             done.record(self._reduce_stream)
             bucket.reduced_event = done
@@ -248,8 +260,9 @@ class NaiveDdp(nn.Module):
             # iter would leave it partial — reduce what's there)
             for b in self._buckets:
                 if b.ready > 0:
-                    dist.all_reduce(b.data, op=dist.ReduceOp.AVG,
-                                    group=self.group)
+                    dist.all_reduce(b.data,
+                                    op=dist.ReduceOp.AVG if self._avg()
+                                    else dist.ReduceOp.SUM, group=self.group)
                     b.reset()
             for b in self._buckets:
                 for p, v in zip(b.params, b.views):
@@ -258,13 +271,15 @@ class NaiveDdp(nn.Module):
         else:
             for work, b in self._works:
                 work.wait()
-                b.data.div_(self._world())
+                if self._avg():
+                    b.data.div_(self._world())
             self._works.clear()
             for b in self._buckets:
                 if b.ready > 0:
                     dist.all_reduce(b.data, op=dist.ReduceOp.SUM,
                                     group=self.group)
-                    b.data.div_(self._world())
+                    if self._avg():
+                        b.data.div_(self._world())
                     b.reset()
             for b in self._buckets:
                 for p, v in zip(b.params, b.views):
