@@ -39,6 +39,9 @@ def parse_args():
     ap.add_argument("--model", type=str, default="gpt2_1.3b",
                     choices=["gpt2_1.3b", "gpt2_small", "tiny", "llama_8b",
                              "moe_8x"])
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the whole train step in one hipGraph "
+                         "(world_size==1 only)")
     ap.add_argument("--zero", action="store_true",
                     help="use Bf16ZeroOptimizer (hybrid node-local shard) + "
                          "sharded EMA instead of plain FusedAdamW")
@@ -64,7 +67,6 @@ def main():
     from torchdistpackage_amd.ops.optim import FusedAdamW
 
     world = int(os.environ.get("WORLD_SIZE", 1))
-    n_gpus = args.gpus if args.gpus > 1 else world
     if world > 1:
         info = setup_distributed()
         rank = info["rank"]
@@ -209,15 +211,29 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         if ema is not None:
             ema.update()
 
+    stepper = step
+    if args.graph and world == 1 and dev.type == "cuda" and ema is None:
+        # hipGraph capture: grads must keep stable storage
+        from torchdistpackage_amd.utils_graph import GraphedStep
+
+        def graph_step():
+            out = model(x, labels=x)
+            out["loss"].backward()
+            opt.step()
+            opt.zero_grad(set_to_none=False)
+
+        gs = GraphedStep(graph_step, warmup=max(args.warmup, 3))
+        stepper = gs.replay
+
     for _ in range(args.warmup):
-        step()
+        stepper()
     if dist.is_initialized():
         dist.barrier()
     if dev.type == "cuda":
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        stepper()
     if dev.type == "cuda":
         torch.cuda.synchronize()
     if dist.is_initialized():

@@ -13,7 +13,6 @@ import time
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-import torch.nn.functional as F
 
 
 class PPStage(nn.Module):

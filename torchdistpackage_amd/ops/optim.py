@@ -100,10 +100,13 @@ class FusedAdamW(torch.optim.Optimizer):
             beta1, beta2 = group["betas"]
             if fg.mt_ready:
                 from . import ext
-                gptrs = torch.tensor(
-                    [p.grad.data_ptr() if p.grad is not None else 0
-                     for p in fg.params], dtype=torch.int64)
-                gptrs = gptrs.to(fg.master.device, non_blocking=True)
+                key = tuple(p.grad.data_ptr() if p.grad is not None else 0
+                            for p in fg.params)
+                if getattr(fg, "_gptr_key", None) != key:
+                    fg._gptr_key = key
+                    fg._gptr_dev = torch.tensor(
+                        list(key), dtype=torch.int64).to(fg.master.device)
+                gptrs = fg._gptr_dev
                 grad_dtype = next((p.grad.dtype for p in fg.params
                                    if p.grad is not None), fg.uniform_dtype)
                 ext("multi_adamw").multi_adamw_step(

@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ...ops import flash_attention, fused_qkv_attention
+from ...ops import fused_qkv_attention
 from .tp_utils import (ColParallelLinear, RowParallelLinear, TpLinear,
                        copy_to_tp_region, gather_from_sequence_parallel_region,
                        get_tp_size, is_sequence_parallel)

@@ -55,8 +55,6 @@ class TpMlp(nn.Module):
         self.fc2 = RowParallelLinear(dim * hidden_mult, dim, bias=bias,
                                      sequence_parallel=sequence_parallel,
                                      device=device, dtype=dtype)
-        # keep fc1's bias separate so bias+gelu fuse into one kernel pass
-        self._fused_act = True
 
     def forward(self, x):
         from .tp_utils import copy_to_tp_region, \

@@ -237,6 +237,28 @@ class _SplitToSp(torch.autograd.Function):
         return _all_gather_first_dim(grad.contiguous())
 
 
+class _GatherLastDim(torch.autograd.Function):
+    """fwd: all-gather along the LAST dim (ColParallel gather_output);
+    bwd: take the local slice."""
+
+    @staticmethod
+    def forward(ctx, x):
+        tp = get_tp_size()
+        if tp == 1:
+            return x
+        xs = [torch.empty_like(x) for _ in range(tp)]
+        dist.all_gather(xs, x.contiguous(), group=get_tp_group())
+        return torch.cat(xs, dim=-1)
+
+    @staticmethod
+    def backward(ctx, grad):
+        tp = get_tp_size()
+        if tp == 1:
+            return grad
+        n = grad.shape[-1] // tp
+        return grad.narrow(-1, get_tp_rank() * n, n).contiguous()
+
+
 # ---------------------------------------------------------------------------
 # layers
 # ---------------------------------------------------------------------------
@@ -284,7 +306,7 @@ class ColParallelLinear(TpLinear):
         x = copy_to_tp_region(x)
         out = F.linear(x, self.weight, self.bias)
         if self.gather_output and get_tp_size() > 1:
-            raise NotImplementedError("gather_output not used in this stack")
+            out = _GatherLastDim.apply(out)
         return out
 
     @torch.no_grad()
