@@ -135,6 +135,11 @@ def main():
                 "global_batch": B_global,
                 "seq_len": args.seq,
                 "parallelism": f"dp{dp}_pp{pp}_tp{tp}",
+                **({"torchddp_ms_per_step":
+                    round(result["torchddp_ms_per_step"], 3),
+                    "naive_vs_torchddp_speedup":
+                    round(result["naive_vs_torchddp_speedup"], 4)}
+                   if "torchddp_ms_per_step" in result else {}),
             },
         }), flush=True)
     if world > 1:
@@ -244,7 +249,52 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         t = torch.tensor([dt], device=dev if dev.type == "cuda" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dt = float(t.item())
-    return {"ms_per_step": dt * 1e3}
+    result = {"ms_per_step": dt * 1e3}
+
+    # --- NaiveDdp vs TorchDDP step time (headline sub-metric) ----------
+    # best-effort: a failure here must never kill the main bench result
+    try:
+        _compare_torchddp(args, model, dev, dp, x, result)
+    except Exception as e:  # noqa: BLE001
+        print(f"[bench] torchddp comparison skipped: {e}", file=sys.stderr)
+    return result
+
+
+def _compare_torchddp(args, model, dev, dp, x, result):
+    import time
+    import torch.distributed as dist
+    from torchdistpackage_amd import tpc
+    from torchdistpackage_amd.ddp import NaiveDdp
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+    if isinstance(model, NaiveDdp) and dp > 1 and dev.type == "cuda" \
+            and args.model.startswith("gpt2"):
+        import torch.nn as nn
+        model.remove_hooks()
+        tddp = nn.parallel.DistributedDataParallel(
+            model.module, process_group=tpc.get_group("data"))
+        opt2 = FusedAdamW(tddp.parameters(), lr=1e-4, weight_decay=0.1)
+
+        def step2():
+            out = tddp(x, labels=x)
+            out["loss"].backward()
+            opt2.step()
+            opt2.zero_grad()
+
+        for _ in range(args.warmup):
+            step2()
+        dist.barrier()
+        torch.cuda.synchronize()
+        t0b = time.perf_counter()
+        for _ in range(args.steps):
+            step2()
+        torch.cuda.synchronize()
+        dist.barrier()
+        dtb = (time.perf_counter() - t0b) / args.steps
+        tb = torch.tensor([dtb], device=dev)
+        dist.all_reduce(tb, op=dist.ReduceOp.MAX)
+        result["torchddp_ms_per_step"] = float(tb.item()) * 1e3
+        result["naive_vs_torchddp_speedup"] = \
+            result["torchddp_ms_per_step"] / result["ms_per_step"]
 
 
 if __name__ == "__main__":

@@ -286,6 +286,12 @@ class NaiveDdp(nn.Module):
                     if p.grad is not None:
                         p.grad.copy_(v)
 
+    def remove_hooks(self):
+        """Detach all grad hooks (e.g. to re-wrap the module elsewhere)."""
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()
+
     def zero_grad(self, set_to_none: bool = True):
         self.module.zero_grad(set_to_none=set_to_none)
 
