@@ -32,9 +32,11 @@ class GPT2Config:
     dim: int = 768
     max_seq: int = 1024
     hidden_mult: int = 4
+    dropout: float = 0.0
     causal: bool = True
     sequence_parallel: bool = True
     tie_weights: bool = True
+    checkpoint_activations: bool = False  # recompute blocks in backward
 
 
 def gpt2_small() -> GPT2Config:
@@ -96,7 +98,7 @@ class GPT2Model(nn.Module):
         self.embed = GPT2Embedding(cfg, **kw)
         self.blocks = nn.ModuleList([
             ParallelBlock(cfg.dim, cfg.n_head, cfg.hidden_mult,
-                          causal=cfg.causal,
+                          causal=cfg.causal, dropout=cfg.dropout,
                           sequence_parallel=cfg.sequence_parallel, **kw)
             for _ in range(cfg.n_layer)])
         self.head = GPT2Head(cfg, self.embed.wte if cfg.tie_weights else None,
@@ -105,8 +107,14 @@ class GPT2Model(nn.Module):
     def forward(self, idx: torch.Tensor,
                 labels: Optional[torch.Tensor] = None) -> dict:
         x = self.embed(idx)
+        use_ckpt = self.cfg.checkpoint_activations and self.training \
+            and torch.is_grad_enabled()
         for blk in self.blocks:
-            x = blk(x)
+            if use_ckpt:
+                from torch.utils.checkpoint import checkpoint
+                x = checkpoint(blk, x, use_reentrant=False)
+            else:
+                x = blk(x)
         from ..parallel.tensor import (is_sequence_parallel,
                                        gather_from_sequence_parallel_region)
         if is_sequence_parallel(x) and get_tp_size() > 1:

@@ -48,8 +48,10 @@ class TpMlp(nn.Module):
     """
 
     def __init__(self, dim: int, hidden_mult: int = 4, bias: bool = True,
-                 sequence_parallel: bool = False, device=None, dtype=None):
+                 dropout: float = 0.0, sequence_parallel: bool = False,
+                 device=None, dtype=None):
         super().__init__()
+        self.dropout = dropout
         self.fc1 = ColParallelLinear(dim, dim * hidden_mult, bias=bias,
                                      device=device, dtype=dtype)
         self.fc2 = RowParallelLinear(dim * hidden_mult, dim, bias=bias,
@@ -67,6 +69,8 @@ class TpMlp(nn.Module):
             x = copy_to_tp_region(x)
         h = F.linear(x, self.fc1.weight)  # bias deferred to fused kernel
         h = bias_gelu(h, self.fc1.bias)
+        if self.dropout > 0 and self.training:
+            h = F.dropout(h, p=self.dropout)
         return self.fc2(h)
 
     @torch.no_grad()

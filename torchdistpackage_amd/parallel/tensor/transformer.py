@@ -53,16 +53,17 @@ class ParallelBlock(nn.Module):
     """
 
     def __init__(self, dim: int, n_head: int, hidden_mult: int = 4,
-                 bias: bool = True, causal: bool = True,
+                 bias: bool = True, causal: bool = True, dropout: float = 0.0,
                  sequence_parallel: bool = True, device=None, dtype=None):
         super().__init__()
         kw = {"device": device, "dtype": dtype}
         self.sequence_parallel = sequence_parallel and get_tp_size() > 1
+        self.dropout = dropout
         self.ln_1 = LayerNorm(dim, **kw)
         self.attn = TpAttention(dim, n_head, bias=bias, causal=causal,
                                 sequence_parallel=self.sequence_parallel, **kw)
         self.ln_2 = LayerNorm(dim, **kw)
-        self.mlp = TpMlp(dim, hidden_mult, bias=bias,
+        self.mlp = TpMlp(dim, hidden_mult, bias=bias, dropout=dropout,
                          sequence_parallel=self.sequence_parallel, **kw)
         if self.sequence_parallel:
             # LN grads come from the local sequence shard -> all-reduce over
@@ -77,7 +78,11 @@ class ParallelBlock(nn.Module):
         h = self.ln_1(x)
         if self.sequence_parallel:
             set_sequence_parallel_attr(h)
-        x = x + self.attn(h)
+        a = self.attn(h)
+        if self.dropout > 0 and self.training:
+            import torch.nn.functional as F
+            a = F.dropout(a, p=self.dropout)
+        x = x + a
         h = self.ln_2(x)
         if self.sequence_parallel:
             set_sequence_parallel_attr(h)
