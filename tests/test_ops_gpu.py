@@ -38,6 +38,17 @@ def test_mfma_layout_probe():
         f"got:\n{d.cpu()[:4, :4]}\nref:\n{ref[:4, :4]}"
 
 
+def test_mfma_layout_probe_32():
+    """32x32x16 fragment-layout assumption used by attention v2."""
+    torch.manual_seed(1)
+    a = (torch.randn(32, 16) * 0.5).bfloat16().to(_dev())
+    b = (torch.randn(16, 32) * 0.5).bfloat16().to(_dev())
+    d = ops.ext("probe").mfma_probe_32x32x16(a, b)
+    ref = a.float().cpu() @ b.float().cpu()
+    err = (d.cpu() - ref).abs().max().item()
+    assert err < 0.05, f"32x32 MFMA layout mismatch: {err}"
+
+
 # ---------------------------------------------------------------- norms
 
 @pytest.mark.parametrize("shape", [(4, 128, 1024), (2, 63, 2048), (1, 8, 64)])

@@ -21,6 +21,7 @@ SOURCES = [
     "norms.hip",
     "elementwise.hip",
     "attention.hip",
+    "attention_v2.hip",
     "cross_entropy.hip",
     "probe.hip",
 ]

@@ -26,6 +26,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -604,6 +605,10 @@ Strides get_strides(const torch::Tensor& t) {
 
 }  // namespace
 
+void attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                 torch::Tensor o, torch::Tensor lse, bool causal,
+                 double scale);
+
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
                                     bool causal, double scale) {
@@ -616,6 +621,11 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
   TORCH_CHECK(k.size(2) == S, "cross-attention S_kv != S_q not supported yet");
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat));
+  static const bool force_v1 = std::getenv("TDPA_ATTN_V1") != nullptr;
+  if (D == 128 && !force_v1) {
+    attn_fwd_v2(q, k, v, o, lse, causal, scale);
+    return {o, lse};
+  }
   auto stream = at::cuda::getCurrentHIPStream();
   constexpr int QT = NW_FWD * WQ;
   dim3 grid((S + QT - 1) / QT, B * H), block(NW_FWD * 64);

@@ -35,6 +35,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor dv,
                                     bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
+torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);
@@ -54,6 +55,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
+  m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
 }

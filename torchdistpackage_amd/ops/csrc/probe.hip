@@ -48,3 +48,49 @@ torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b) {
   HIP_CHECK_LAST();
   return d;
 }
+
+namespace {
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// 32x32x16 layout probe: A[32][16] @ B[16][32] with the assumed maps
+//   A: lane l holds A[row=l&31][k=(l>>5)*8+j]  (8 bf16)
+//   B: lane l holds B[k=(l>>5)*8+j][col=l&31]
+//   C/D: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)   (16 f32)
+__global__ void mfma_probe32_kernel(const unsigned short* __restrict__ a,
+                                    const unsigned short* __restrict__ b,
+                                    float* __restrict__ d) {
+  const int lane = threadIdx.x & 63;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+  unsigned short af[8], bf[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[l31 * 16 + hi * 8 + j];
+    bf[j] = b[(hi * 8 + j) * 32 + l31];
+  }
+  f32x16 acc = (f32x16)(0.f);
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+      *(const bf16x8_v*)af, *(const bf16x8_v*)bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    d[row * 32 + l31] = acc[r];
+  }
+}
+
+}  // namespace
+
+torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({32, 16}));
+  TORCH_CHECK(b.sizes() == torch::IntArrayRef({16, 32}));
+  auto d = torch::empty({32, 32}, a.options().dtype(torch::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, stream,
+                     (const unsigned short*)a.contiguous().data_ptr(),
+                     (const unsigned short*)b.contiguous().data_ptr(),
+                     d.data_ptr<float>());
+  HIP_CHECK_LAST();
+  return d;
+}
