@@ -608,6 +608,13 @@ Strides get_strides(const torch::Tensor& t) {
 void attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor o, torch::Tensor lse, bool causal,
                  double scale);
+void attn_bwd_dq_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                    torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                    torch::Tensor dq, bool causal, double scale);
+void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                      torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                      torch::Tensor dk, torch::Tensor dv, bool causal,
+                      double scale);
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
@@ -679,6 +686,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                        delta.data_ptr<float>(), os, dos, B, H, S, D);
   }
 
+  static const bool force_v1b = std::getenv("TDPA_ATTN_V1") != nullptr;
+  if (D == 128 && !force_v1b) {
+    attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
+    attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
+    HIP_CHECK_LAST();
+    return {dq, dk, dv};
+  }
   constexpr int QT_DQ = NW_DQ * WQ;
   constexpr int QT_KV = NW_DKDV * WQ;
   dim3 grid_dq((S + QT_DQ - 1) / QT_DQ, B * H), block_dq(NW_DQ * 64);

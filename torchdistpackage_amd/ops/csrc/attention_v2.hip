@@ -305,3 +305,439 @@ void attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        B, H, S, (float)scale, q_per_kv);
   HIP_CHECK_LAST();
 }
+
+// ===========================================================================
+// Backward v2 (D=128): same swapped-operand structure as the forward.
+//
+// dq pass: wave owns 32 q-rows (lane = one q-column of the S^T tiles);
+//   per 64-key tile: S^T = mfma(K, Q^T), dP^T = mfma(V, dO^T); P and dS stay
+//   in registers (lse/delta are per-lane scalars); dq^T += mfma(K^T, dS^T)
+//   with the dS^T B-fragments built by the cvt_pk+permlane exchange.
+// ===========================================================================
+
+namespace {
+
+template <bool CAUSAL>
+__launch_bounds__(512)
+__global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
+                                      const unsigned short* __restrict__ k,
+                                      const unsigned short* __restrict__ v,
+                                      const unsigned short* __restrict__ dout,
+                                      const float* __restrict__ lse,
+                                      const float* __restrict__ delta,
+                                      unsigned short* __restrict__ dq,
+                                      Strides2 qs, Strides2 ks, Strides2 vs,
+                                      Strides2 dos, Strides2 dqs,
+                                      int B, int H, int S, float scale,
+                                      int q_per_kv) {
+  __shared__ __attribute__((aligned(16))) unsigned short k_lds[KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short v_lds[KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short kt_lds[D2 * KV];
+
+  const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
+  const int hkv = hh / q_per_kv;
+  const int qbase = blockIdx.x * QT2;
+  if (qbase >= S) return;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const unsigned short* qp = q + bb * qs.b + hh * qs.h;
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
+  const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
+
+  const int qrow = qbase + wid * QW + l31;
+  bf16x8_v q_frag[8], do_frag[8];
+  float my_lse, my_delta;
+  {
+    long r = (long)(qrow < S ? qrow : S - 1) * qs.s;
+    long rdo = (long)(qrow < S ? qrow : S - 1) * dos.s;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      q_frag[c] = pack8v(qp + r + c * 16 + hi * 8);
+      do_frag[c] = pack8v(dop + rdo + c * 16 + hi * 8);
+    }
+    my_lse = qrow < S ? lse[(long)bh * S + qrow] : 0.f;
+    my_delta = qrow < S ? delta[(long)bh * S + qrow] : 0.f;
+  }
+
+  f32x16 acc[4];  // dq^T[d][q]
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) acc[ds] = (f32x16)(0.f);
+
+  const int kv_end = CAUSAL ? min(S, qbase + QT2) : S;
+
+  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
+    __syncthreads();
+    // direct global->LDS staging (no prefetch registers: this kernel sits
+    // at the 256-VGPR edge and prefetch regs push it into scratch spills)
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int key = idx / D2;
+      int col = idx % D2;
+      int gkey = kt0 + key;
+      bf16x8_v kv8, vv8;
+      if (gkey < S) {
+        kv8 = pack8v(kp + (long)gkey * ks.s + col);
+        vv8 = pack8v(vp + (long)gkey * vs.s + col);
+      } else {
+        kv8 = (bf16x8_v)(__bf16)0.f;
+        vv8 = (bf16x8_v)(__bf16)0.f;
+      }
+      *(bf16x8_v*)&k_lds[swzK(key, col)] = kv8;
+      *(bf16x8_v*)&v_lds[swzK(key, col)] = vv8;
+      const unsigned short* ksrc = (const unsigned short*)&kv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kt_lds[swzV(col + j, key)] = ksrc[j];
+    }
+    __syncthreads();
+
+    // S^T and dP^T for the 2 key-subtiles
+    f32x16 st[2], dpt[2];
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      st[kt] = (f32x16)(0.f);
+      dpt[kt] = (f32x16)(0.f);
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        bf16x8_v a_k = pack8v(&k_lds[swzK(kt * 32 + l31, c * 16 + hi * 8)]);
+        st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_k, q_frag[c], st[kt], 0, 0, 0);
+        bf16x8_v a_v = pack8v(&v_lds[swzK(kt * 32 + l31, c * 16 + hi * 8)]);
+        dpt[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_v, do_frag[c], dpt[kt], 0, 0, 0);
+      }
+    }
+
+    // dS^T = P * (dP^T - delta) * scale, P = exp(S^T*scale - lse)
+    // (written back into st[] — fresh arrays would push the kernel past the
+    // 256-VGPR budget and spill)
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int key = kt0 + kt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
+        float pv = valid ? __expf(st[kt][r] * scale - my_lse) : 0.f;
+        st[kt][r] = pv * (dpt[kt][r] - my_delta) * scale;
+      }
+    }
+
+    // dS^T B-fragments (same exchange as the forward)
+    bf16x8_v db[2][2];
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        int m0 = 2 * kc, m1 = 2 * kc + 1;
+        unsigned a0 = cvt_pk_bf16(st[kt][4 * m0], st[kt][4 * m0 + 1]);
+        unsigned a1 = cvt_pk_bf16(st[kt][4 * m0 + 2], st[kt][4 * m0 + 3]);
+        unsigned b0 = cvt_pk_bf16(st[kt][4 * m1], st[kt][4 * m1 + 1]);
+        unsigned b1 = cvt_pk_bf16(st[kt][4 * m1 + 2], st[kt][4 * m1 + 3]);
+        auto s0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+        auto s1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+        unsigned f0, f1, f2, f3;
+        if (hi == 0) {
+          f0 = a0; f1 = a1; f2 = s0[1]; f3 = s1[1];
+        } else {
+          f0 = s0[0]; f1 = s1[0]; f2 = b0; f3 = b1;
+        }
+        unsigned* dst = (unsigned*)&db[kt][kc];
+        dst[0] = f0; dst[1] = f1; dst[2] = f2; dst[3] = f3;
+      }
+    }
+
+    // dq^T += K^T @ dS^T
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8_v a_kt = pack8v(&kt_lds[swzV(ds * 32 + l31,
+                                              kt * 32 + kc * 16 + hi * 8)]);
+          acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_kt, db[kt][kc], acc[ds], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  if (qrow < S) {
+    unsigned short* dqp = dq + bb * dqs.b + hh * dqs.h + (long)qrow * dqs.s;
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int blk = 0; blk < 4; ++blk) {
+        int d0 = ds * 32 + 8 * blk + 4 * hi;
+        unsigned short out4[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          out4[j] = f2bf(acc[ds][4 * blk + j]);
+        *(ushort4*)(dqp + d0) = *(ushort4*)out4;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void attn_bwd_dq_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                    torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                    torch::Tensor dq, bool causal, double scale) {
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  const int q_per_kv = H / Hkv;
+  TORCH_CHECK(D == 128);
+  auto get = [](const torch::Tensor& t) {
+    return Strides2{t.stride(0), t.stride(1), t.stride(2)};
+  };
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
+#define L_DQ2(CC)                                                             \
+  hipLaunchKernelGGL((attn_bwd_dq_v2_kernel<CC>), grid, block, 0, stream,     \
+                     (const unsigned short*)q.data_ptr(),                     \
+                     (const unsigned short*)k.data_ptr(),                     \
+                     (const unsigned short*)v.data_ptr(),                     \
+                     (const unsigned short*)dout.data_ptr(),                  \
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),          \
+                     (unsigned short*)dq.data_ptr(), get(q), get(k), get(v),  \
+                     get(dout), get(dq), B, H, S, (float)scale, q_per_kv)
+  if (causal) L_DQ2(true); else L_DQ2(false);
+#undef L_DQ2
+  HIP_CHECK_LAST();
+}
+
+// ===========================================================================
+// dkdv v2: two passes (dv, then dk), each with the wave owning 32 KEYS
+// (lane = one key-column).  A single fused kernel needs both 64-register
+// accumulators plus K^T AND V^T fragments and spills ~240 B/lane; splitting
+// re-runs the QK^T MFMAs (cheap) but keeps both kernels spill-free.
+//   dv pass: S = mfma(Q, K^T regs); P = exp(S*scale - lse[q]);
+//            dv^T += mfma(dO^T, P-frag)
+//   dk pass: S as above + dP = mfma(dO, V^T regs);
+//            dS = P*(dP - delta[q])*scale; dk^T += mfma(Q^T, dS-frag)
+// GQA: outputs are per-q-head partials (summed by the host wrapper).
+// ===========================================================================
+
+namespace {
+
+template <bool CAUSAL, bool DK_PASS>
+__launch_bounds__(512)
+__global__ void attn_bwd_dkdv_v2_kernel(
+    const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse,
+    const float* __restrict__ delta,
+    unsigned short* __restrict__ out,   // dk (DK_PASS) or dv
+    Strides2 qs, Strides2 ks, Strides2 vs, Strides2 dos, Strides2 outs,
+    int B, int H, int S, float scale, int q_per_kv) {
+  // LDS: Q and dO tiles (A-operands), plus the transposed tile the pass's
+  // final MFMA consumes (dO^T for dv, Q^T for dk)
+  __shared__ __attribute__((aligned(16))) unsigned short q_lds[KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short do_lds[KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short tr_lds[D2 * KV];
+  __shared__ float lse_lds[KV];
+  __shared__ float del_lds[KV];
+
+  const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
+  const int hkv = hh / q_per_kv;
+  const int kbase = blockIdx.x * QT2;
+  if (kbase >= S) return;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const unsigned short* qp = q + bb * qs.b + hh * qs.h;
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
+  const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
+  const float* lsep = lse + (long)bh * S;
+  const float* delp = delta + (long)bh * S;
+
+  // this wave's 32 keys: K^T fragments always; V^T only in the dk pass
+  const int key = kbase + wid * QW + l31;
+  bf16x8_v kt_frag[8];
+  bf16x8_v vt_frag[DK_PASS ? 8 : 1];
+  {
+    long r = (long)(key < S ? key : S - 1);
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      kt_frag[c] = pack8v(kp + r * ks.s + c * 16 + hi * 8);
+      if (DK_PASS)
+        vt_frag[c] = pack8v(vp + r * vs.s + c * 16 + hi * 8);
+    }
+  }
+
+  f32x16 acc[4];  // dk^T or dv^T (col = key)
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) acc[ds] = (f32x16)(0.f);
+
+  const int qt_start = CAUSAL ? kbase : 0;
+
+  for (int qt0 = qt_start; qt0 < S; qt0 += KV) {
+    __syncthreads();
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int row = idx / D2;
+      int col = idx % D2;
+      int grow = qt0 + row;
+      bf16x8_v qv, dv8;
+      if (grow < S) {
+        qv = pack8v(qp + (long)grow * qs.s + col);
+        dv8 = pack8v(dop + (long)grow * dos.s + col);
+      } else {
+        qv = (bf16x8_v)(__bf16)0.f;
+        dv8 = (bf16x8_v)(__bf16)0.f;
+      }
+      *(bf16x8_v*)&q_lds[swzK(row, col)] = qv;
+      if (DK_PASS)
+        *(bf16x8_v*)&do_lds[swzK(row, col)] = dv8;
+      // transposed tile: dO^T for the dv pass, Q^T for the dk pass
+      const unsigned short* tsrc = DK_PASS
+          ? (const unsigned short*)&qv : (const unsigned short*)&dv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        tr_lds[swzV(col + j, row)] = tsrc[j];
+    }
+    if (tid < KV) {
+      int grow = qt0 + tid;
+      lse_lds[tid] = grow < S ? lsep[grow] : 0.f;
+      if (DK_PASS)
+        del_lds[tid] = grow < S ? delp[grow] : 0.f;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int qt = 0; qt < 2; ++qt) {
+      f32x16 s_acc = (f32x16)(0.f);
+      f32x16 dp_acc;
+      if (DK_PASS) dp_acc = (f32x16)(0.f);
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        bf16x8_v a_q = pack8v(&q_lds[swzK(qt * 32 + l31, c * 16 + hi * 8)]);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            a_q, kt_frag[c], s_acc, 0, 0, 0);
+        if (DK_PASS) {
+          bf16x8_v a_do = pack8v(
+              &do_lds[swzK(qt * 32 + l31, c * 16 + hi * 8)]);
+          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_do, vt_frag[c], dp_acc, 0, 0, 0);
+        }
+      }
+
+      // P into s_acc (dv pass) or dS into s_acc (dk pass)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qrel = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        int qrow = qt0 + qrel;
+        bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
+        float p = valid ?
+            __expf(s_acc[r] * scale - lse_lds[qrel]) : 0.f;
+        s_acc[r] = DK_PASS
+            ? p * (dp_acc[r] - del_lds[qrel]) * scale : p;
+      }
+
+      // B-fragments over the q dimension (exchange as in the forward)
+      bf16x8_v fb[2];
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        int m0 = 2 * kc, m1 = 2 * kc + 1;
+        unsigned a0 = cvt_pk_bf16(s_acc[4 * m0], s_acc[4 * m0 + 1]);
+        unsigned a1 = cvt_pk_bf16(s_acc[4 * m0 + 2], s_acc[4 * m0 + 3]);
+        unsigned b0 = cvt_pk_bf16(s_acc[4 * m1], s_acc[4 * m1 + 1]);
+        unsigned b1 = cvt_pk_bf16(s_acc[4 * m1 + 2], s_acc[4 * m1 + 3]);
+        auto s0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+        auto s1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+        unsigned* dst = (unsigned*)&fb[kc];
+        if (hi == 0) {
+          dst[0] = a0; dst[1] = a1; dst[2] = s0[1]; dst[3] = s1[1];
+        } else {
+          dst[0] = s0[0]; dst[1] = s1[0]; dst[2] = b0; dst[3] = b1;
+        }
+      }
+
+      // dv^T += dO^T @ P   or   dk^T += Q^T @ dS
+#pragma unroll
+      for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8_v a_tr = pack8v(&tr_lds[swzV(ds * 32 + l31,
+                                              qt * 32 + kc * 16 + hi * 8)]);
+          acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_tr, fb[kc], acc[ds], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  if (key < S) {
+    unsigned short* op = out + bb * outs.b + hh * outs.h +
+                         (long)key * outs.s;
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int blk = 0; blk < 4; ++blk) {
+        int d0 = ds * 32 + 8 * blk + 4 * hi;
+        unsigned short o4[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          o4[j] = f2bf(acc[ds][4 * blk + j]);
+        *(ushort4*)(op + d0) = *(ushort4*)o4;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                      torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                      torch::Tensor dk, torch::Tensor dv, bool causal,
+                      double scale) {
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  const int q_per_kv = H / Hkv;
+  TORCH_CHECK(D == 128);
+  auto get = [](const torch::Tensor& t) {
+    return Strides2{t.stride(0), t.stride(1), t.stride(2)};
+  };
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
+#define L_KV2(CC)                                                             \
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false>), grid, block, 0,  \
+                       stream, (const unsigned short*)q.data_ptr(),           \
+                       (const unsigned short*)k.data_ptr(),                   \
+                       (const unsigned short*)v.data_ptr(),                   \
+                       (const unsigned short*)dout.data_ptr(),                \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                       (unsigned short*)dv.data_ptr(), get(q), get(k),        \
+                       get(v), get(dout), get(dv), B, H, S, (float)scale,     \
+                       q_per_kv);                                             \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true>), grid, block, 0,   \
+                       stream, (const unsigned short*)q.data_ptr(),           \
+                       (const unsigned short*)k.data_ptr(),                   \
+                       (const unsigned short*)v.data_ptr(),                   \
+                       (const unsigned short*)dout.data_ptr(),                \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),        \
+                       (unsigned short*)dk.data_ptr(), get(q), get(k),        \
+                       get(v), get(dout), get(dk), B, H, S, (float)scale,     \
+                       q_per_kv);                                             \
+  } while (0)
+  if (causal) L_KV2(true); else L_KV2(false);
+#undef L_KV2
+  HIP_CHECK_LAST();
+}
