@@ -67,8 +67,10 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
                                    Strides2 os,
                                    int B, int H, int S, float scale,
                                    int q_per_kv) {
-  __shared__ __attribute__((aligned(16))) unsigned short k_lds[KV * D2];
-  __shared__ __attribute__((aligned(16))) unsigned short vt_lds[D2 * KV];
+  // double-buffered tiles: occupancy is VGPR-bound (1 block/CU at 230
+  // VGPRs), so the extra LDS is free and buys one barrier per tile
+  __shared__ __attribute__((aligned(16))) unsigned short k_lds[2][KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short vt_lds[2][D2 * KV];
 
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
@@ -123,23 +125,25 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
     }
   };
   stage_load(0);
-
-  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
-    __syncthreads();
+  // prologue: fill buffer 0
+  {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
       int idx = tid * 8 + c * 4096;
       int key = idx / D2;
       int col = idx % D2;
-      *(bf16x8_v*)&k_lds[swzK(key, col)] = k_reg[c];
+      *(bf16x8_v*)&k_lds[0][swzK(key, col)] = k_reg[c];
       const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        vt_lds[swzV(col + j, key)] = vsrc[j];
+        vt_lds[0][swzV(col + j, key)] = vsrc[j];
     }
+    if (KV < kv_end) stage_load(KV);
     __syncthreads();
-    if (kt0 + KV < kv_end) stage_load(kt0 + KV);
+  }
 
+  int buf = 0;
+  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
     // ---- swapped QK^T: S^T[key][q] for 2 key-subtiles of 32
     f32x16 st[2];
 #pragma unroll
@@ -148,7 +152,8 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
         // A = K[key = kt*32 + l31][k-chunk c]
-        bf16x8_v a_k = pack8v(&k_lds[swzK(kt * 32 + l31, c * 16 + hi * 8)]);
+        bf16x8_v a_k = pack8v(
+            &k_lds[buf][swzK(kt * 32 + l31, c * 16 + hi * 8)]);
         st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             a_k, q_frag[c], st[kt], 0, 0, 0);
       }
@@ -242,13 +247,32 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
       for (int kt = 0; kt < 2; ++kt) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8_v a_v = pack8v(&vt_lds[swzV(ds * 32 + l31,
-                                             kt * 32 + kc * 16 + hi * 8)]);
+          bf16x8_v a_v = pack8v(&vt_lds[buf][swzV(ds * 32 + l31,
+                                                  kt * 32 + kc * 16 + hi * 8)]);
           acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_v, pb[kt][kc], acc[ds], 0, 0, 0);
         }
       }
     }
+
+    // ---- write the prefetched NEXT tile into the other buffer (nobody
+    // reads it until after the barrier) and issue the tile-after-next loads
+    if (kt0 + KV < kv_end) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        int idx = tid * 8 + c * 4096;
+        int key = idx / D2;
+        int col = idx % D2;
+        *(bf16x8_v*)&k_lds[buf ^ 1][swzK(key, col)] = k_reg[c];
+        const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vt_lds[buf ^ 1][swzV(col + j, key)] = vsrc[j];
+      }
+      if (kt0 + 2 * KV < kv_end) stage_load(kt0 + 2 * KV);
+    }
+    __syncthreads();
+    buf ^= 1;
   }
 
   // ---- epilogue: O[q][d] = O^T / l; LSE
