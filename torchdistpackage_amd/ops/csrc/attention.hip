@@ -242,26 +242,31 @@ __global__ void attn_delta_kernel(const unsigned short* __restrict__ o,
                                   float* __restrict__ delta,
                                   Strides os, Strides ds,
                                   int B, int H, int S, int D) {
-  // one wave per row (grid.x covers B*H*S rows), vectorized 8-wide
-  long row = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
-  if (row >= (long)B * H * S) return;
+  // one wave per row, GRID-STRIDED (a one-row-per-block launch was
+  // dispatch-bound: 65k tiny blocks cost 3x the memory floor)
+  const long rows = (long)B * H * S;
+  const int wpb = blockDim.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  int s = row % S;
-  int h = (row / S) % H;
-  int b = row / ((long)S * H);
-  const unsigned short* orow = o + b * os.b + h * os.h + (long)s * os.s;
-  const unsigned short* drow = dout + b * ds.b + h * ds.h + (long)s * ds.s;
-  float acc = 0.f;
-  for (int i = lane * 8; i < D; i += WAVE * 8) {
-    bf16x8_v ov = pack8(orow + i);
-    bf16x8_v dv = pack8(drow + i);
-    const unsigned short* op = (const unsigned short*)&ov;
-    const unsigned short* dp = (const unsigned short*)&dv;
+  const long stride = (long)gridDim.x * wpb;
+  for (long row = (long)blockIdx.x * wpb + threadIdx.x / WAVE; row < rows;
+       row += stride) {
+    int s = row % S;
+    int h = (row / S) % H;
+    int b = row / ((long)S * H);
+    const unsigned short* orow = o + b * os.b + h * os.h + (long)s * os.s;
+    const unsigned short* drow = dout + b * ds.b + h * ds.h + (long)s * ds.s;
+    float acc = 0.f;
+    for (int i = lane * 8; i < D; i += WAVE * 8) {
+      bf16x8_v ov = pack8(orow + i);
+      bf16x8_v dv = pack8(drow + i);
+      const unsigned short* op = (const unsigned short*)&ov;
+      const unsigned short* dp = (const unsigned short*)&dv;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc += bf2f(op[j]) * bf2f(dp[j]);
+      for (int j = 0; j < 8; ++j) acc += bf2f(op[j]) * bf2f(dp[j]);
+    }
+    acc = wave_sum(acc);
+    if (lane == 0) delta[row] = acc;
   }
-  acc = wave_sum(acc);
-  if (lane == 0) delta[row] = acc;
 }
 
 template <int D, bool CAUSAL, int NW>
@@ -678,8 +683,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
 
   {
     long rows = (long)B * H * S;
-    int wpb = 4;
-    long blocks = (rows + wpb - 1) / wpb;
+    long blocks = (rows + 3) / 4;
+    if (blocks > 2048) blocks = 2048;   // grid-stride the rest
     hipLaunchKernelGGL(attn_delta_kernel, dim3((unsigned)blocks), dim3(256),
                        0, stream, (const unsigned short*)o.data_ptr(),
                        (const unsigned short*)dout.data_ptr(),
