@@ -354,9 +354,9 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
                                       Strides2 dos, Strides2 dqs,
                                       int B, int H, int S, float scale,
                                       int q_per_kv) {
-  __shared__ __attribute__((aligned(16))) unsigned short k_lds[KV * D2];
-  __shared__ __attribute__((aligned(16))) unsigned short v_lds[KV * D2];
-  __shared__ __attribute__((aligned(16))) unsigned short kt_lds[D2 * KV];
+  __shared__ __attribute__((aligned(16))) unsigned short k_lds[2][KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short v_lds[2][KV * D2];
+  __shared__ __attribute__((aligned(16))) unsigned short kt_lds[2][D2 * KV];
 
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
@@ -395,10 +395,10 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
 
   const int kv_end = CAUSAL ? min(S, qbase + QT2) : S;
 
-  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
-    __syncthreads();
-    // direct global->LDS staging (no prefetch registers: this kernel sits
-    // at the 256-VGPR edge and prefetch regs push it into scratch spills)
+  // double-buffered: stage tile t+1 while nobody reads that buffer, one
+  // barrier per tile (direct load->write; no prefetch registers: this
+  // kernel sits at the 256-VGPR edge)
+  auto stage_dq = [&](int kt0, int b) {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
       int idx = tid * 8 + c * 4096;
@@ -413,14 +413,19 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
         kv8 = (bf16x8_v)(__bf16)0.f;
         vv8 = (bf16x8_v)(__bf16)0.f;
       }
-      *(bf16x8_v*)&k_lds[swzK(key, col)] = kv8;
-      *(bf16x8_v*)&v_lds[swzK(key, col)] = vv8;
+      *(bf16x8_v*)&k_lds[b][swzK(key, col)] = kv8;
+      *(bf16x8_v*)&v_lds[b][swzK(key, col)] = vv8;
       const unsigned short* ksrc = (const unsigned short*)&kv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        kt_lds[swzV(col + j, key)] = ksrc[j];
+        kt_lds[b][swzV(col + j, key)] = ksrc[j];
     }
-    __syncthreads();
+  };
+  stage_dq(0, 0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
 
     // S^T and dP^T for the 2 key-subtiles
     f32x16 st[2], dpt[2];
@@ -430,10 +435,12 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
       dpt[kt] = (f32x16)(0.f);
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        bf16x8_v a_k = pack8v(&k_lds[swzK(kt * 32 + l31, c * 16 + hi * 8)]);
+        bf16x8_v a_k = pack8v(
+            &k_lds[buf][swzK(kt * 32 + l31, c * 16 + hi * 8)]);
         st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             a_k, q_frag[c], st[kt], 0, 0, 0);
-        bf16x8_v a_v = pack8v(&v_lds[swzK(kt * 32 + l31, c * 16 + hi * 8)]);
+        bf16x8_v a_v = pack8v(
+            &v_lds[buf][swzK(kt * 32 + l31, c * 16 + hi * 8)]);
         dpt[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             a_v, do_frag[c], dpt[kt], 0, 0, 0);
       }
@@ -484,13 +491,18 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
       for (int kt = 0; kt < 2; ++kt) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8_v a_kt = pack8v(&kt_lds[swzV(ds * 32 + l31,
-                                              kt * 32 + kc * 16 + hi * 8)]);
+          bf16x8_v a_kt = pack8v(&kt_lds[buf][swzV(ds * 32 + l31,
+                                                   kt * 32 + kc * 16 +
+                                                   hi * 8)]);
           acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_kt, db[kt][kc], acc[ds], 0, 0, 0);
         }
       }
     }
+
+    if (kt0 + KV < kv_end) stage_dq(kt0 + KV, buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
   }
 
   if (qrow < S) {
@@ -566,11 +578,13 @@ __global__ void attn_bwd_dkdv_v2_kernel(
     int B, int H, int S, float scale, int q_per_kv) {
   // LDS: Q and dO tiles (A-operands), plus the transposed tile the pass's
   // final MFMA consumes (dO^T for dv, Q^T for dk)
-  __shared__ __attribute__((aligned(16))) unsigned short q_lds[KV * D2];
-  __shared__ __attribute__((aligned(16))) unsigned short do_lds[KV * D2];
-  __shared__ __attribute__((aligned(16))) unsigned short tr_lds[D2 * KV];
-  __shared__ float lse_lds[KV];
-  __shared__ float del_lds[KV];
+  __shared__ __attribute__((aligned(16))) unsigned short q_lds[2][KV * D2];
+  // dO tile only exists in the dk pass (dv reads dO^T via tr_lds)
+  __shared__ __attribute__((aligned(16)))
+      unsigned short do_lds[DK_PASS ? 2 * KV * D2 : 8];
+  __shared__ __attribute__((aligned(16))) unsigned short tr_lds[2][D2 * KV];
+  __shared__ float lse_lds[2][KV];
+  __shared__ float del_lds[2][KV];
 
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
@@ -610,8 +624,7 @@ __global__ void attn_bwd_dkdv_v2_kernel(
 
   const int qt_start = CAUSAL ? kbase : 0;
 
-  for (int qt0 = qt_start; qt0 < S; qt0 += KV) {
-    __syncthreads();
+  auto stage_kv = [&](int qt0, int b) {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
       int idx = tid * 8 + c * 4096;
@@ -626,23 +639,28 @@ __global__ void attn_bwd_dkdv_v2_kernel(
         qv = (bf16x8_v)(__bf16)0.f;
         dv8 = (bf16x8_v)(__bf16)0.f;
       }
-      *(bf16x8_v*)&q_lds[swzK(row, col)] = qv;
+      *(bf16x8_v*)&q_lds[b][swzK(row, col)] = qv;
       if (DK_PASS)
-        *(bf16x8_v*)&do_lds[swzK(row, col)] = dv8;
+        *(bf16x8_v*)&do_lds[b * KV * D2 + swzK(row, col)] = dv8;
       // transposed tile: dO^T for the dv pass, Q^T for the dk pass
       const unsigned short* tsrc = DK_PASS
           ? (const unsigned short*)&qv : (const unsigned short*)&dv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        tr_lds[swzV(col + j, row)] = tsrc[j];
+        tr_lds[b][swzV(col + j, row)] = tsrc[j];
     }
     if (tid < KV) {
       int grow = qt0 + tid;
-      lse_lds[tid] = grow < S ? lsep[grow] : 0.f;
+      lse_lds[b][tid] = grow < S ? lsep[grow] : 0.f;
       if (DK_PASS)
-        del_lds[tid] = grow < S ? delp[grow] : 0.f;
+        del_lds[b][tid] = grow < S ? delp[grow] : 0.f;
     }
-    __syncthreads();
+  };
+  stage_kv(qt_start, 0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int qt0 = qt_start; qt0 < S; qt0 += KV) {
 
 #pragma unroll
     for (int qt = 0; qt < 2; ++qt) {
@@ -651,12 +669,13 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       if (DK_PASS) dp_acc = (f32x16)(0.f);
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        bf16x8_v a_q = pack8v(&q_lds[swzK(qt * 32 + l31, c * 16 + hi * 8)]);
+        bf16x8_v a_q = pack8v(
+            &q_lds[buf][swzK(qt * 32 + l31, c * 16 + hi * 8)]);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             a_q, kt_frag[c], s_acc, 0, 0, 0);
         if (DK_PASS) {
           bf16x8_v a_do = pack8v(
-              &do_lds[swzK(qt * 32 + l31, c * 16 + hi * 8)]);
+              &do_lds[buf * KV * D2 + swzK(qt * 32 + l31, c * 16 + hi * 8)]);
           dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_do, vt_frag[c], dp_acc, 0, 0, 0);
         }
@@ -669,9 +688,9 @@ __global__ void attn_bwd_dkdv_v2_kernel(
         int qrow = qt0 + qrel;
         bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
         float p = valid ?
-            __expf(s_acc[r] * scale - lse_lds[qrel]) : 0.f;
+            __expf(s_acc[r] * scale - lse_lds[buf][qrel]) : 0.f;
         s_acc[r] = DK_PASS
-            ? p * (dp_acc[r] - del_lds[qrel]) * scale : p;
+            ? p * (dp_acc[r] - del_lds[buf][qrel]) * scale : p;
       }
 
       // B-fragments over the q dimension (exchange as in the forward)
@@ -698,13 +717,18 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       for (int ds = 0; ds < 4; ++ds) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8_v a_tr = pack8v(&tr_lds[swzV(ds * 32 + l31,
-                                              qt * 32 + kc * 16 + hi * 8)]);
+          bf16x8_v a_tr = pack8v(&tr_lds[buf][swzV(ds * 32 + l31,
+                                                   qt * 32 + kc * 16 +
+                                                   hi * 8)]);
           acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_tr, fb[kc], acc[ds], 0, 0, 0);
         }
       }
     }
+
+    if (qt0 + KV < S) stage_kv(qt0 + KV, buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
   }
 
   if (key < S) {
