@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import math
 import os
-from typing import List, Optional
+from typing import Optional
 
 import torch
 

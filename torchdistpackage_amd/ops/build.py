@@ -11,7 +11,7 @@ Usage: ``python -m torchdistpackage_amd.ops.build`` or
 from __future__ import annotations
 
 import os
-import sys
+
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
