@@ -5,3 +5,4 @@ from .tensor import (TpLinear, ColParallelLinear, RowParallelLinear,
 from .pipeline import (forward_backward, forward_eval, partition_uniform,
                        partition_balanced, flatten_model, flatten_sequence,
                        flat_and_partition, clip_grad_norm_, NativeScalerPP)
+from .context import ulysses_attention, UlyssesAttention, ring_attention

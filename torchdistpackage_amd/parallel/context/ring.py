@@ -1,0 +1,181 @@
+"""Ring attention: context parallelism with K/V blocks rotating the ring.
+
+The long-context strategy the reference only sketches as blockwise math
+(explore/flash-attn/tile_attn.py "is the correct starting spec
+... for ring-attention partials since it maintains running max/expsum").
+Built directly on the in-tree flash kernels:
+
+- forward: cp ring steps; each step computes a flash partial (O_b, LSE_b) of
+  the local Q against the currently-held K/V block; partials merge with the
+  stable rule  lse = logaddexp(lse_a, lse_b);
+  o = o_a*exp(lse_a-lse) + o_b*exp(lse_b-lse).
+- backward: the exact blockwise-gradient identity (reference
+  tile_attn.py:156-212): re-run the per-block flash BACKWARD with the GLOBAL
+  (merged) LSE and the global delta = rowsum(do*o) — each block's dq
+  contribution sums locally, while (dk, dv) accumulators travel around the
+  ring with the K/V blocks and arrive back at their owners after a full
+  rotation.  The per-block backward is the same gfx950 kernel set the local
+  path uses (it takes lse/delta as inputs).
+
+Causal convention: rank r owns contiguous sequence block r; block r attends
+blocks < r fully and itself causally.  (A zigzag shard layout balances the
+causal work better and drops in by re-indexing shards.)
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from ...ops import ext, flash_attention
+
+
+def _shift(x: torch.Tensor, group, direction: int) -> torch.Tensor:
+    """Rotate a tensor one step around the group ring (blocking)."""
+    ranks = dist.get_process_group_ranks(group)
+    n = len(ranks)
+    me = ranks.index(dist.get_rank())
+    dst = ranks[(me + direction) % n]
+    src = ranks[(me - direction) % n]
+    out = torch.empty_like(x)
+    ops = [dist.P2POp(dist.isend, x.contiguous(), dst),
+           dist.P2POp(dist.irecv, out, src)]
+    for r in dist.batch_isend_irecv(ops):
+        r.wait()
+    if x.is_cuda:
+        torch.cuda.synchronize()
+    return out
+
+
+def _fwd_partial(q, k, v, causal, scale):
+    """(o, lse) of one block, no autograd."""
+    if q.is_cuda:
+        o = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        _, lse = ext("flash_attention").attn_fwd(q, k, v, o, bool(causal),
+                                                 float(scale))
+        return o, lse
+    qf, kf, vf = q.float(), k.float(), v.float()
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        S = q.shape[-2]
+        mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
+                          device=q.device).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    o = torch.matmul(torch.softmax(s, -1), vf).to(q.dtype)
+    return o, lse
+
+
+def _bwd_partial(do, q, k, v, o, lse, causal, scale):
+    """(dq, dk, dv) of one block given the GLOBAL lse (and o for delta)."""
+    if q.is_cuda:
+        dq = torch.empty_like(q)
+        dk = torch.empty_like(q)   # per-q-head partials (H == Hkv here)
+        dv = torch.empty_like(q)
+        ext("flash_attention").attn_bwd(do, q, k, v, o, lse, dq, dk, dv,
+                                        bool(causal), float(scale))
+        return dq, dk, dv
+    qf, kf, vf = q.float(), k.float(), v.float()
+    dof, of = do.float(), o.float()
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        S = q.shape[-2]
+        mask = torch.ones(S, k.shape[-2], dtype=torch.bool,
+                          device=q.device).tril_()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = torch.exp(s - lse.unsqueeze(-1).float())
+    dv = torch.matmul(p.transpose(-1, -2), dof)
+    dp = torch.matmul(dof, vf.transpose(-1, -2))
+    delta = (dof * of).sum(-1, keepdim=True)
+    ds = p * (dp - delta) * scale
+    dq = torch.matmul(ds, kf)
+    dk = torch.matmul(ds.transpose(-1, -2), qf)
+    return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype)
+
+
+class _RingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale, group_holder):
+        group = group_holder[0]
+        cp = dist.get_world_size(group)
+        ranks = dist.get_process_group_ranks(group)
+        me = ranks.index(dist.get_rank())
+
+        o_acc = None
+        lse_acc = None
+        k_cur, v_cur = k.contiguous(), v.contiguous()
+        with torch.no_grad():
+            for step in range(cp):
+                src = (me - step) % cp
+                attend = (src < me) or (src == me) or (not causal)
+                if attend:
+                    o_p, lse_p = _fwd_partial(
+                        q, k_cur, v_cur, causal and src == me, scale)
+                    if o_acc is None:
+                        o_acc, lse_acc = o_p, lse_p
+                    else:
+                        new_lse = torch.logaddexp(lse_acc, lse_p)
+                        w_a = torch.exp(lse_acc - new_lse) \
+                            .unsqueeze(-1).to(o_p.dtype)
+                        w_b = torch.exp(lse_p - new_lse) \
+                            .unsqueeze(-1).to(o_p.dtype)
+                        o_acc = o_acc * w_a + o_p * w_b
+                        lse_acc = new_lse
+                if step + 1 < cp:
+                    kv = torch.stack([k_cur, v_cur])
+                    kv = _shift(kv, group, +1)
+                    k_cur, v_cur = kv[0].contiguous(), kv[1].contiguous()
+        ctx.save_for_backward(q, k, v, o_acc, lse_acc)
+        ctx.meta = (causal, scale, group)
+        return o_acc
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        causal, scale, group = ctx.meta
+        cp = dist.get_world_size(group)
+        ranks = dist.get_process_group_ranks(group)
+        me = ranks.index(dist.get_rank())
+        do = do.contiguous()
+
+        dq_acc = torch.zeros_like(q, dtype=torch.float32)
+        # traveling accumulators ride WITH the K/V blocks (fp32)
+        dk_acc = torch.zeros_like(k, dtype=torch.float32)
+        dv_acc = torch.zeros_like(v, dtype=torch.float32)
+        k_cur, v_cur = k.contiguous(), v.contiguous()
+        for step in range(cp):
+            src = (me - step) % cp
+            attend = (src < me) or (src == me) or (not causal)
+            if attend:
+                dq_p, dk_p, dv_p = _bwd_partial(
+                    do, q, k_cur, v_cur, o, lse,
+                    causal and src == me, scale)
+                dq_acc += dq_p.float()
+                dk_acc += dk_p.float()
+                dv_acc += dv_p.float()
+            # full rotation (cp shifts) returns accumulators to owners
+            blob = torch.stack([k_cur.float(), v_cur.float(),
+                                dk_acc, dv_acc])
+            blob = _shift(blob, group, +1)
+            k_cur = blob[0].to(k.dtype).contiguous()
+            v_cur = blob[1].to(v.dtype).contiguous()
+            dk_acc = blob[2].contiguous()
+            dv_acc = blob[3].contiguous()
+        return (dq_acc.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype),
+                None, None, None)
+
+
+def ring_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   causal: bool = True, scale: Optional[float] = None,
+                   group: Optional[dist.ProcessGroup] = None):
+    """q/k/v: (B, H, S_local, D) — CP-group rank r owns sequence block r.
+    Returns the (B, H, S_local, D) output shard with exact gradients."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if group is None or not dist.is_initialized() \
+            or dist.get_world_size(group) == 1:
+        return flash_attention(q, k, v, causal=causal, scale=scale)
+    return _RingAttention.apply(q, k, v, causal, scale, (group,))
