@@ -37,7 +37,7 @@ class MoEConfig:
 
 def mixtral_style_8x() -> MoEConfig:
     """8-expert GPT-2-medium-class MoE (per-token params ~ dense medium)."""
-    return MoEConfig(n_layer=24, n_head=16, dim=1024, num_experts=8, top_k=2)
+    return MoEConfig(n_layer=24, n_head=8, dim=1024, num_experts=8, top_k=2)
 
 
 class MoEBlock(nn.Module):

@@ -211,29 +211,29 @@ class ExpertParallelMoE(nn.Module):
         received = _AllToAll.apply(dispatched, in_splits, out_splits,
                                    self.ep_group)
 
-        # received tokens are ordered [src_rank][local_expert]; regroup per
-        # local expert across src ranks
-        outs = torch.empty_like(received)
-        # offsets of each (src, local_expert) segment in `received`
+        # received tokens are ordered [src_rank][local_expert]; ONE stable
+        # argsort by local-expert id groups them contiguously (the per-expert
+        # cat/copy loop this replaces cost ~7k small copyBuffer launches per
+        # MoE bench step)
         seg_sizes = my_slice.reshape(-1).cpu()    # (ep * num_local,)
-        seg_offs = torch.cumsum(
-            torch.cat([torch.zeros(1, dtype=seg_sizes.dtype), seg_sizes]),
-            0).tolist()
+        seg_expert = torch.arange(self.ep_size * self.num_local) \
+            % self.num_local
+        tok_expert = torch.repeat_interleave(seg_expert, seg_sizes) \
+            .to(received.device)
+        order2 = torch.argsort(tok_expert, stable=True)
+        grouped = received[order2]
+        per_expert = torch.bincount(tok_expert.cpu(),
+                                    minlength=self.num_local).tolist()
+        y_parts = []
+        off = 0
         for le in range(self.num_local):
-            idxs = []
-            for src in range(self.ep_size):
-                seg = src * self.num_local + le
-                s, e = int(seg_offs[seg]), int(seg_offs[seg + 1])
-                if e > s:
-                    idxs.append((s, e))
-            if not idxs:
-                continue
-            chunk = torch.cat([received[s:e] for s, e in idxs], dim=0)
-            y = self.experts[le](chunk)
-            off = 0
-            for s, e in idxs:
-                outs[s:e] = y[off:off + (e - s)]
-                off += e - s
+            n = per_expert[le]
+            if n > 0:
+                y_parts.append(self.experts[le](grouped[off:off + n]))
+            off += n
+        y_all = torch.cat(y_parts, dim=0) if y_parts else grouped[:0]
+        outs = torch.empty_like(received)
+        outs[order2] = y_all
 
         returned = _AllToAll.apply(outs, out_splits, in_splits, self.ep_group)
 

@@ -80,7 +80,12 @@ class FusedAdamW(torch.optim.Optimizer):
     def _build(self):
         for group in self.param_groups:
             ps = [p for p in group["params"] if p.requires_grad]
-            group["_flat"] = _FlatGroup(ps) if ps else None
+            # split by dtype so each flat stays multi-tensor-kernel eligible
+            # (e.g. fp32 MoE router gates among bf16 params)
+            by_dtype = {}
+            for p in ps:
+                by_dtype.setdefault(p.dtype, []).append(p)
+            group["_flats"] = [_FlatGroup(v) for v in by_dtype.values()]
         self._built = True
 
     @torch.no_grad()
@@ -93,9 +98,13 @@ class FusedAdamW(torch.optim.Optimizer):
             self._build()
         self._step += 1
         for group in self.param_groups:
-            fg: _FlatGroup = group.get("_flat")
-            if fg is None:
-                continue
+            for fg in group.get("_flats", []):
+                self._step_flat(group, fg)
+        return loss
+
+    @torch.no_grad()
+    def _step_flat(self, group, fg: _FlatGroup):
+        if True:
             lr = group["lr"]
             beta1, beta2 = group["betas"]
             if fg.mt_ready:
@@ -123,7 +132,6 @@ class FusedAdamW(torch.optim.Optimizer):
                              self._step, lr, beta1, beta2, group["eps"],
                              group["weight_decay"])
                 torch._foreach_copy_(fg.params, fg.master_views)
-        return loss
 
     def zero_grad(self, set_to_none: bool = True):
         for group in self.param_groups:
@@ -139,9 +147,9 @@ class FusedAdamW(torch.optim.Optimizer):
         return {
             "step": self._step,
             "groups": [
-                {"master": g["_flat"].master if g["_flat"] else None,
-                 "exp_avg": g["_flat"].exp_avg if g["_flat"] else None,
-                 "exp_avg_sq": g["_flat"].exp_avg_sq if g["_flat"] else None}
+                [{"master": fg.master, "exp_avg": fg.exp_avg,
+                  "exp_avg_sq": fg.exp_avg_sq}
+                 for fg in g.get("_flats", [])]
                 for g in self.param_groups],
         }
 
@@ -150,11 +158,9 @@ class FusedAdamW(torch.optim.Optimizer):
         if not self._built:
             self._build()
         self._step = sd["step"]
-        for g, gsd in zip(self.param_groups, sd["groups"]):
-            fg = g.get("_flat")
-            if fg is None or gsd["master"] is None:
-                continue
-            fg.master.copy_(gsd["master"])
-            fg.exp_avg.copy_(gsd["exp_avg"])
-            fg.exp_avg_sq.copy_(gsd["exp_avg_sq"])
-            torch._foreach_copy_(fg.params, fg.master_views)
+        for g, gsds in zip(self.param_groups, sd["groups"]):
+            for fg, fsd in zip(g.get("_flats", []), gsds):
+                fg.master.copy_(fsd["master"])
+                fg.exp_avg.copy_(fsd["exp_avg"])
+                fg.exp_avg_sq.copy_(fsd["exp_avg_sq"])
+                torch._foreach_copy_(fg.params, fg.master_views)
