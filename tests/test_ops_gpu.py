@@ -250,7 +250,7 @@ def test_fused_adamw_multi_tensor():
             t.addcdiv_(m, (v / bc2).sqrt().add_(1e-8), value=-1e-2 / bc1)
         opt.zero_grad()
 
-    fg = opt.param_groups[0]["_flat"]
+    fg = opt.param_groups[0]["_flats"][0]
     assert fg.mt_ready, "multi-tensor path must be active on GPU"
     off = 0
     for p, ref in zip(model.parameters(), cpu_master):

@@ -224,12 +224,24 @@ class ExpertParallelMoE(nn.Module):
         grouped = received[order2]
         per_expert = torch.bincount(tok_expert.cpu(),
                                     minlength=self.num_local).tolist()
+        # Quantize each expert's batch to a multiple of 1024 (zero-padded,
+        # padding sliced off the output).  Routing drifts every step, and
+        # every UNSEEN (M,N,K) costs a host-side hipBLASLt heuristic pass
+        # (~5-10 ms): unquantized expert GEMMs made deep MoE forwards
+        # host-bound (measured 90 ms/block cold vs 2 ms warm).
+        Q = 1024
         y_parts = []
         off = 0
         for le in range(self.num_local):
             n = per_expert[le]
             if n > 0:
-                y_parts.append(self.experts[le](grouped[off:off + n]))
+                seg = grouped[off:off + n]
+                npad = (n + Q - 1) // Q * Q
+                if npad != n:
+                    pad = torch.zeros(npad - n, seg.shape[1],
+                                      dtype=seg.dtype, device=seg.device)
+                    seg = torch.cat([seg, pad], dim=0)
+                y_parts.append(self.experts[le](seg)[:n])
             off += n
         y_all = torch.cat(y_parts, dim=0) if y_parts else grouped[:0]
         outs = torch.empty_like(received)
