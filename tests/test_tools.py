@@ -121,3 +121,30 @@ def test_partition_by_time_smoke():
     parts = partition_by_time(layers, 2, torch.randn(16, 32), warmup=1,
                               iters=2)
     assert len(parts) == 2 and parts[0][0] == 0 and parts[-1][1] == 4
+
+
+def test_bias_gelu_no_bias_cpu():
+    import torchdistpackage_amd.ops as ops
+    x = torch.randn(4, 16, requires_grad=True)
+    y = ops.bias_gelu(x, None)
+    ref = torch.nn.functional.gelu(x.detach(), approximate="tanh")
+    assert torch.allclose(y, ref, atol=1e-6)
+    y.sum().backward()
+    assert x.grad is not None
+
+
+def test_rope_rotation_property():
+    """RoPE must preserve norms and give relative-position-dependent dots."""
+    from torchdistpackage_amd.models.llama import Rope
+    rope = Rope(64, 128, 10000.0)
+    x = torch.randn(1, 1, 16, 64)
+    y = rope(x)
+    assert torch.allclose(x.norm(dim=-1), y.norm(dim=-1), atol=1e-5)
+    # dot of rotated q,k at positions (i, j) depends only on i-j
+    q = torch.randn(64)
+    k = torch.randn(64)
+    def rot(v, pos):
+        return rope(v.view(1, 1, 1, 64).expand(1, 1, 128, 64))[0, 0, pos]
+    d1 = torch.dot(rot(q, 3), rot(k, 5))
+    d2 = torch.dot(rot(q, 10), rot(k, 12))
+    assert torch.allclose(d1, d2, atol=1e-4)
