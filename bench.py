@@ -171,27 +171,25 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         model = GPT2Model(cfg, device=dev, dtype=dtype)
 
     ema = None
-    if world > 1 and dp > 1:
-        model = NaiveDdp(model, group=tpc.get_group("data"))
-        if args.model == "moe_8x":
-            from torchdistpackage_amd.ddp import create_moe_dp_hooks
-            inner = model.module
-            if tpc.get_group_size("moe_dp") > 1:
-                create_moe_dp_hooks(list(inner.expert_parameters()))
     if args.zero:
+        # ZeRO owns grad reduction — model stays unwrapped (wrapping with
+        # NaiveDdp too would leave BOTH hook sets firing)
         from torchdistpackage_amd import Bf16ZeroOptimizer, ShardedEMA, \
             setup_node_groups
-        inner_model = model.module if isinstance(model, NaiveDdp) else model
         node_group = setup_node_groups(num_per_node=8) if world > 1 else None
-        inner_opt = FusedAdamW(inner_model.parameters(), lr=1e-4,
-                               weight_decay=0.1)
+        inner_opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
         opt = Bf16ZeroOptimizer(inner_opt, group=node_group,
                                 grad_group=tpc.get_group("data")
                                 if world > 1 else None, stage2=True)
-        ema = ShardedEMA(inner_model, decay=0.999)
-        # ZeRO owns grad reduction: unhook NaiveDdp if wrapped
-        model = inner_model
+        ema = ShardedEMA(model, decay=0.999)
     else:
+        if world > 1 and dp > 1:
+            model = NaiveDdp(model, group=tpc.get_group("data"))
+            if args.model == "moe_8x":
+                from torchdistpackage_amd.ddp import create_moe_dp_hooks
+                inner = model.module
+                if tpc.get_group_size("moe_dp") > 1:
+                    create_moe_dp_hooks(list(inner.expert_parameters()))
         opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
 
     # identical data inside a TP group; different across DP ranks

@@ -96,12 +96,10 @@ def run_pp_bench(args, cfg, dev, dtype, dp, pp, tp):
     stage.labels_full = x
 
     def fwd_fn(stage_in):
-        inp = stage_in
-        if stage.is_first:
-            # scheduler passes the sliced token micro-batch
-            pass
-        return stage_ddp(inp) if isinstance(stage_ddp, NaiveDdp) \
-            else stage(inp)
+        # first stage receives the sliced token micro-batch, later stages
+        # the previous stage's activation shard
+        return stage_ddp(stage_in) if isinstance(stage_ddp, NaiveDdp) \
+            else stage(stage_in)
 
     def step():
         stage.reset_iter()
