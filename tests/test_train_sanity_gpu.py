@@ -1,0 +1,71 @@
+"""GPU training-sanity tests: loss must fall on a repeated synthetic batch
+for every model family (validates fwd+bwd+optimizer end-to-end on the HIP
+kernel path, not just per-op numerics)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _overfit(model, x, steps=15, lr=3e-4):
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+    opt = FusedAdamW(model.parameters(), lr=lr)
+    first = None
+    for it in range(steps):
+        loss = model(x, labels=x)["loss"]
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        if it == 0:
+            first = loss.item()
+    last = loss.item()
+    assert last == last and first == first, "nan loss"
+    assert last < first * 0.9, f"loss did not fall: {first} -> {last}"
+    return first, last
+
+
+def test_gpt2_overfits():
+    from torchdistpackage_amd.models.gpt2 import GPT2Config, GPT2Model
+    torch.manual_seed(0)
+    cfg = GPT2Config(vocab_size=1024, n_layer=3, n_head=4, dim=256,
+                     max_seq=128)
+    m = GPT2Model(cfg, device="cuda", dtype=torch.bfloat16)
+    _overfit(m, torch.randint(0, 1024, (4, 128), device="cuda"))
+
+
+def test_llama_overfits():
+    from torchdistpackage_amd.models.llama import LlamaModel, llama_tiny
+    torch.manual_seed(0)
+    m = LlamaModel(llama_tiny(), device="cuda", dtype=torch.bfloat16)
+    _overfit(m, torch.randint(0, 512, (4, 128), device="cuda"))
+
+
+def test_moe_overfits():
+    from torchdistpackage_amd.models.moe_model import MoEConfig, MoEModel
+    torch.manual_seed(0)
+    cfg = MoEConfig(vocab_size=512, n_layer=2, n_head=2, dim=256, max_seq=64,
+                    num_experts=4, top_k=2, hidden_mult=2)
+    m = MoEModel(cfg, device="cuda", dtype=torch.bfloat16)
+    _overfit(m, torch.randint(0, 512, (4, 64), device="cuda"))
+
+
+def test_zero_optimizer_trains_gpu():
+    """Bf16ZeroOptimizer single-rank GPU path (fp32 master + fused inner)."""
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+    from torchdistpackage_amd.models.gpt2 import GPT2Config, GPT2Model
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+    torch.manual_seed(0)
+    cfg = GPT2Config(vocab_size=512, n_layer=2, n_head=4, dim=128, max_seq=64)
+    m = GPT2Model(cfg, device="cuda", dtype=torch.bfloat16)
+    opt = Bf16ZeroOptimizer(FusedAdamW(m.parameters(), lr=3e-4))
+    x = torch.randint(0, 512, (4, 64), device="cuda")
+    first = None
+    for it in range(10):
+        loss = m(x, labels=x)["loss"]
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        if it == 0:
+            first = loss.item()
+    assert loss.item() < first
