@@ -56,7 +56,7 @@ def test_zero_optimizer_trains_gpu():
     from torchdistpackage_amd.models.gpt2 import GPT2Config, GPT2Model
     from torchdistpackage_amd.ops.optim import FusedAdamW
     torch.manual_seed(0)
-    cfg = GPT2Config(vocab_size=512, n_layer=2, n_head=4, dim=128, max_seq=64)
+    cfg = GPT2Config(vocab_size=512, n_layer=2, n_head=2, dim=128, max_seq=64)
     m = GPT2Model(cfg, device="cuda", dtype=torch.bfloat16)
     opt = Bf16ZeroOptimizer(FusedAdamW(m.parameters(), lr=3e-4))
     x = torch.randint(0, 512, (4, 64), device="cuda")
