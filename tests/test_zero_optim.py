@@ -96,3 +96,34 @@ def _zero_clip(rank, world_size):
 
 def test_zero_clip_grad():
     run_distributed(_zero_clip, world_size=2)
+
+
+def _zero_with_fused_adamw(rank, world_size):
+    """Composition regression: ZeRO's master views have requires_grad=False
+    and must still be updated by FusedAdamW (a silent no-op bug once)."""
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+    from torchdistpackage_amd.ops.optim import FusedAdamW
+
+    model = _make_model(seed=21)
+    opt = Bf16ZeroOptimizer(FusedAdamW(model.parameters(), lr=1e-2))
+    p0 = next(model.parameters()).detach().clone()
+    losses = []
+    torch.manual_seed(3)
+    x = torch.randn(4, 48)
+    for it in range(5):
+        loss = model(x).pow(2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(loss.item())
+    assert not torch.equal(p0, next(model.parameters())), "params frozen"
+    assert losses[-1] < losses[0], f"loss flat: {losses}"
+    return True
+
+
+def test_zero_with_fused_adamw():
+    run_distributed(_zero_with_fused_adamw, world_size=1)
+
+
+def test_zero_with_fused_adamw_world2():
+    run_distributed(_zero_with_fused_adamw, world_size=2)

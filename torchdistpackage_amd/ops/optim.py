@@ -79,7 +79,10 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def _build(self):
         for group in self.param_groups:
-            ps = [p for p in group["params"] if p.requires_grad]
+            # NOTE: no requires_grad filter — like stock torch optimizers.
+            # (Bf16ZeroOptimizer rebinds groups to fp32 master VIEWS whose
+            # requires_grad is False; their .grad is assigned manually.)
+            ps = list(group["params"])
             # split by dtype so each flat stays multi-tensor-kernel eligible
             # (e.g. fp32 MoE router gates among bf16 params)
             by_dtype = {}
