@@ -72,15 +72,16 @@ class ShardedEMA:
         d = self.decay if decay is None else decay
         if self._flat.numel() == 0:
             return
-        # flat_param snapshot of this shard's params, then one fused lerp:
-        # ema = d*ema + (1-d)*p  ==  ema.lerp_(p, 1-d)
-        src = torch.cat([self._params[i].detach().reshape(-1).float()
-                         for i in self._my_idx])
-        try:
-            from ..ops import ema_update_
-            ema_update_(self._flat, src, d)
-        except Exception:
-            self._flat.lerp_(src, 1.0 - d)
+        # per-param view updates: a single flat cat would materialize a full
+        # fp32 copy of the shard (~30 GB on Llama-8B) — OOM territory
+        for i in self._my_idx:
+            p = self._params[i].detach().reshape(-1)
+            v = self._views[i]
+            if p.dtype == torch.float32 and v.is_cuda:
+                from ..ops import ema_update_
+                ema_update_(v, p, d)
+            else:
+                v.lerp_(p.float(), 1.0 - d)
 
     @torch.no_grad()
     def state_dict_shard(self) -> Dict[str, torch.Tensor]:

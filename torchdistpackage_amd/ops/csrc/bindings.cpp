@@ -18,8 +18,9 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 double beta2, double eps, double wd);
 void multi_adamw_step(torch::Tensor cpid, torch::Tensor coff,
                       torch::Tensor pptrs, torch::Tensor gptrs,
-                      torch::Tensor moffs, torch::Tensor numels,
-                      torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                      torch::Tensor mptrs, torch::Tensor moffs,
+                      torch::Tensor numels,
+                      torch::Tensor m, torch::Tensor v,
                       long step, double lr, double beta1, double beta2,
                       double eps, double wd, bool param_bf16, bool grad_bf16);
 void ema_update(torch::Tensor ema, torch::Tensor p, double decay);

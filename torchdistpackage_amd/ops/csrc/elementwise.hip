@@ -308,10 +308,10 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
                                    const long* __restrict__ coff,
                                    const long* __restrict__ pptrs,
                                    const long* __restrict__ gptrs,
+                                   const long* __restrict__ mptrs,
                                    const long* __restrict__ moffs,
                                    const long* __restrict__ numels,
                                    int nchunks,
-                                   float* __restrict__ master,
                                    float* __restrict__ m,
                                    float* __restrict__ v,
                                    float lr, float beta1, float beta2,
@@ -323,7 +323,8 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
   const long off = coff[c];
   const long n = numels[pid];
   const long end = off + MT_CHUNK < n ? off + MT_CHUNK : n;
-  const long mbase = moffs[pid];
+  const long mvbase = moffs[pid];     // m/v flat offset
+  float* master = (float*)mptrs[pid];  // per-param master (MAY alias param)
   void* pp = (void*)pptrs[pid];
   const void* gp = (const void*)gptrs[pid];
   for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
@@ -331,16 +332,15 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
     if (gp == nullptr) gi = 0.f;
     else if (grad_bf16) gi = bf2f(((const unsigned short*)gp)[i]);
     else gi = ((const float*)gp)[i];
-    const long k = mbase + i;
-    float pi = master[k] * (1.f - lr * wd);
+    const long k = mvbase + i;
+    float pi = master[i] * (1.f - lr * wd);
     float mi = m[k] * beta1 + gi * (1.f - beta1);
     float vi = v[k] * beta2 + gi * gi * (1.f - beta2);
     m[k] = mi;
     v[k] = vi;
     pi = pi - lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
-    master[k] = pi;
+    master[i] = pi;
     if (param_bf16) ((unsigned short*)pp)[i] = f2bf(pi);
-    else ((float*)pp)[i] = pi;
   }
 }
 
@@ -348,12 +348,13 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
 
 void multi_adamw_step(torch::Tensor cpid, torch::Tensor coff,
                       torch::Tensor pptrs, torch::Tensor gptrs,
-                      torch::Tensor moffs, torch::Tensor numels,
-                      torch::Tensor master, torch::Tensor m, torch::Tensor v,
+                      torch::Tensor mptrs, torch::Tensor moffs,
+                      torch::Tensor numels,
+                      torch::Tensor m, torch::Tensor v,
                       long step, double lr, double beta1, double beta2,
                       double eps, double wd, bool param_bf16,
                       bool grad_bf16) {
-  TORCH_CHECK(master.is_cuda() && master.scalar_type() == torch::kFloat);
+  TORCH_CHECK(m.is_cuda() && m.scalar_type() == torch::kFloat);
   int nchunks = cpid.numel();
   float bc1 = 1.f - powf((float)beta1, (float)step);
   float bc2 = 1.f - powf((float)beta2, (float)step);
@@ -361,8 +362,9 @@ void multi_adamw_step(torch::Tensor cpid, torch::Tensor coff,
   hipLaunchKernelGGL(multi_adamw_kernel, dim3(nchunks), dim3(BLOCK), 0,
                      stream, cpid.data_ptr<int>(), coff.data_ptr<long>(),
                      pptrs.data_ptr<long>(), gptrs.data_ptr<long>(),
-                     moffs.data_ptr<long>(), numels.data_ptr<long>(),
-                     nchunks, master.data_ptr<float>(), m.data_ptr<float>(),
+                     mptrs.data_ptr<long>(), moffs.data_ptr<long>(),
+                     numels.data_ptr<long>(),
+                     nchunks, m.data_ptr<float>(),
                      v.data_ptr<float>(), (float)lr, (float)beta1,
                      (float)beta2, (float)eps, (float)wd, bc1, bc2,
                      param_bf16 ? 1 : 0, grad_bf16 ? 1 : 0);

@@ -29,29 +29,38 @@ class _FlatGroup:
         self.params = params
         n = sum(p.numel() for p in params)
         dev = params[0].device
-        self.master = torch.empty(n, dtype=torch.float32, device=dev)
-        self.grad32 = torch.empty(n, dtype=torch.float32, device=dev)
+        dtypes = {p.dtype for p in params}
+        self.uniform_dtype = params[0].dtype if len(dtypes) == 1 else None
+        self.mt_ready = dev.type == "cuda" and self.uniform_dtype in (
+            torch.bfloat16, torch.float32)
+        # fp32 params on the multi-tensor path update IN PLACE: no master
+        # copy (the ZeRO composition passes fp32 master VIEWS here — a
+        # duplicate master + grad32 wasted +64 GB on Llama-8B)
+        inplace_fp32 = self.mt_ready and self.uniform_dtype == torch.float32
+        offs = []
+        off = 0
+        for p in params:
+            offs.append(off)
+            off += p.numel()
         self.exp_avg = torch.zeros(n, dtype=torch.float32, device=dev)
         self.exp_avg_sq = torch.zeros(n, dtype=torch.float32, device=dev)
         self.master_views = []
         self.grad_views = []
-        offs = []
-        off = 0
-        for p in params:
-            k = p.numel()
-            offs.append(off)
-            mv = self.master.narrow(0, off, k).view_as(p)
-            mv.copy_(p.detach().to(torch.float32))
-            self.master_views.append(mv)
-            self.grad_views.append(self.grad32.narrow(0, off, k).view_as(p))
-            off += k
-        dtypes = {p.dtype for p in params}
-        self.uniform_dtype = params[0].dtype if len(dtypes) == 1 else None
-        # multi-tensor chunk table (GPU, uniform dtype only): one kernel
-        # launch per step, bf16 grads read + bf16 params written in-pass
-        self.mt_ready = False
-        if dev.type == "cuda" and self.uniform_dtype in (
-                torch.bfloat16, torch.float32):
+        if not inplace_fp32:
+            self.master = torch.empty(n, dtype=torch.float32, device=dev)
+            for p, o in zip(params, offs):
+                mv = self.master.narrow(0, o, p.numel()).view_as(p)
+                mv.copy_(p.detach().to(torch.float32))
+                self.master_views.append(mv)
+        else:
+            self.master = None
+            self.master_views = [p for p in params]  # aliases
+        if not self.mt_ready:
+            self.grad32 = torch.empty(n, dtype=torch.float32, device=dev)
+            for p, o in zip(params, offs):
+                self.grad_views.append(
+                    self.grad32.narrow(0, o, p.numel()).view_as(p))
+        if self.mt_ready:
             cpid, coff = [], []
             for i, p in enumerate(params):
                 for c0 in range(0, p.numel(), self.CHUNK):
@@ -61,10 +70,16 @@ class _FlatGroup:
             self.coff = torch.tensor(coff, dtype=torch.int64, device=dev)
             self.pptrs = torch.tensor([p.data_ptr() for p in params],
                                       dtype=torch.int64, device=dev)
+            if inplace_fp32:
+                self.mptrs = self.pptrs
+            else:
+                base = self.master.data_ptr()
+                self.mptrs = torch.tensor(
+                    [base + o * 4 for o in offs], dtype=torch.int64,
+                    device=dev)
             self.moffs = torch.tensor(offs, dtype=torch.int64, device=dev)
             self.numels = torch.tensor([p.numel() for p in params],
                                        dtype=torch.int64, device=dev)
-            self.mt_ready = True
 
 
 class FusedAdamW(torch.optim.Optimizer):
@@ -122,8 +137,8 @@ class FusedAdamW(torch.optim.Optimizer):
                 grad_dtype = next((p.grad.dtype for p in fg.params
                                    if p.grad is not None), fg.uniform_dtype)
                 ext("multi_adamw").multi_adamw_step(
-                    fg.cpid, fg.coff, fg.pptrs, gptrs, fg.moffs, fg.numels,
-                    fg.master, fg.exp_avg, fg.exp_avg_sq, self._step, lr,
+                    fg.cpid, fg.coff, fg.pptrs, gptrs, fg.mptrs, fg.moffs,
+                    fg.numels, fg.exp_avg, fg.exp_avg_sq, self._step, lr,
                     beta1, beta2, group["eps"], group["weight_decay"],
                     fg.uniform_dtype == torch.bfloat16,
                     grad_dtype == torch.bfloat16)
@@ -153,7 +168,7 @@ class FusedAdamW(torch.optim.Optimizer):
                 [{"master": fg.master, "exp_avg": fg.exp_avg,
                   "exp_avg_sq": fg.exp_avg_sq}
                  for fg in g.get("_flats", [])]
-                for g in self.param_groups],
+                for g in self.param_groups],  # master None => in-place fp32
         }
 
     @torch.no_grad()
@@ -163,7 +178,8 @@ class FusedAdamW(torch.optim.Optimizer):
         self._step = sd["step"]
         for g, gsds in zip(self.param_groups, sd["groups"]):
             for fg, fsd in zip(g.get("_flats", []), gsds):
-                fg.master.copy_(fsd["master"])
+                if fg.master is not None and fsd["master"] is not None:
+                    fg.master.copy_(fsd["master"])
+                    torch._foreach_copy_(fg.params, fg.master_views)
                 fg.exp_avg.copy_(fsd["exp_avg"])
                 fg.exp_avg_sq.copy_(fsd["exp_avg_sq"])
-                torch._foreach_copy_(fg.params, fg.master_views)
