@@ -132,7 +132,7 @@ class FusedAdamW(torch.optim.Optimizer):
                 if getattr(fg, "_gptr_key", None) != key:
                     fg._gptr_key = key
                     fg._gptr_dev = torch.tensor(
-                        list(key), dtype=torch.int64).to(fg.master.device)
+                        list(key), dtype=torch.int64).to(fg.exp_avg.device)
                 gptrs = fg._gptr_dev
                 grad_dtype = next((p.grad.dtype for p in fg.params
                                    if p.grad is not None), fg.uniform_dtype)
