@@ -75,7 +75,9 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
   const int hkv = hh / q_per_kv;
-  const int qbase = blockIdx.x * QT2;
+  // causal: schedule heavy (late) q-tiles first so the tail packs evenly
+  const int qtile = CAUSAL ? (gridDim.x - 1 - blockIdx.x) : blockIdx.x;
+  const int qbase = qtile * QT2;
   if (qbase >= S) return;
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;
@@ -361,7 +363,8 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
   const int bh = blockIdx.y;
   const int bb = bh / H, hh = bh % H;
   const int hkv = hh / q_per_kv;
-  const int qbase = blockIdx.x * QT2;
+  const int qtile = CAUSAL ? (gridDim.x - 1 - blockIdx.x) : blockIdx.x;
+  const int qbase = qtile * QT2;
   if (qbase >= S) return;
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;

@@ -122,34 +122,33 @@ class FusedAdamW(torch.optim.Optimizer):
 
     @torch.no_grad()
     def _step_flat(self, group, fg: _FlatGroup):
-        if True:
-            lr = group["lr"]
-            beta1, beta2 = group["betas"]
-            if fg.mt_ready:
-                from . import ext
-                key = tuple(p.grad.data_ptr() if p.grad is not None else 0
-                            for p in fg.params)
-                if getattr(fg, "_gptr_key", None) != key:
-                    fg._gptr_key = key
-                    fg._gptr_dev = torch.tensor(
-                        list(key), dtype=torch.int64).to(fg.exp_avg.device)
-                gptrs = fg._gptr_dev
-                grad_dtype = next((p.grad.dtype for p in fg.params
-                                   if p.grad is not None), fg.uniform_dtype)
-                ext("multi_adamw").multi_adamw_step(
-                    fg.cpid, fg.coff, fg.pptrs, gptrs, fg.mptrs, fg.moffs,
-                    fg.numels, fg.exp_avg, fg.exp_avg_sq, self._step, lr,
-                    beta1, beta2, group["eps"], group["weight_decay"],
-                    fg.uniform_dtype == torch.bfloat16,
-                    grad_dtype == torch.bfloat16)
-            else:
-                grads = [p.grad if p.grad is not None
-                         else torch.zeros_like(p) for p in fg.params]
-                torch._foreach_copy_(fg.grad_views, grads)
-                fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
-                             self._step, lr, beta1, beta2, group["eps"],
-                             group["weight_decay"])
-                torch._foreach_copy_(fg.params, fg.master_views)
+        lr = group["lr"]
+        beta1, beta2 = group["betas"]
+        if fg.mt_ready:
+            from . import ext
+            key = tuple(p.grad.data_ptr() if p.grad is not None else 0
+                        for p in fg.params)
+            if getattr(fg, "_gptr_key", None) != key:
+                fg._gptr_key = key
+                fg._gptr_dev = torch.tensor(
+                    list(key), dtype=torch.int64).to(fg.exp_avg.device)
+            gptrs = fg._gptr_dev
+            grad_dtype = next((p.grad.dtype for p in fg.params
+                               if p.grad is not None), fg.uniform_dtype)
+            ext("multi_adamw").multi_adamw_step(
+                fg.cpid, fg.coff, fg.pptrs, gptrs, fg.mptrs, fg.moffs,
+                fg.numels, fg.exp_avg, fg.exp_avg_sq, self._step, lr,
+                beta1, beta2, group["eps"], group["weight_decay"],
+                fg.uniform_dtype == torch.bfloat16,
+                grad_dtype == torch.bfloat16)
+        else:
+            grads = [p.grad if p.grad is not None
+                     else torch.zeros_like(p) for p in fg.params]
+            torch._foreach_copy_(fg.grad_views, grads)
+            fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
+                         self._step, lr, beta1, beta2, group["eps"],
+                         group["weight_decay"])
+            torch._foreach_copy_(fg.params, fg.master_views)
 
     def zero_grad(self, set_to_none: bool = True):
         for group in self.param_groups:
