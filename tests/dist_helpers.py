@@ -49,19 +49,35 @@ def run_distributed(fn, world_size: int = 2, backend: str = "gloo",
     spawned processes; returns the list of per-rank return values."""
     kwargs = kwargs or {}
     from torchdistpackage_amd.dist.launch import find_free_port
-    port = find_free_port()
-    with tempfile.TemporaryDirectory() as result_dir:
-        ctx = mp.spawn(
-            _worker,
-            args=(world_size, port, fn, args, kwargs, backend, result_dir),
-            nprocs=world_size, join=True, daemon=False)
-        del ctx
-        results = []
-        for r in range(world_size):
-            path = os.path.join(result_dir, f"rank{r}.pkl")
-            assert os.path.exists(path), f"rank {r} produced no result"
-            with open(path, "rb") as f:
-                status, payload = pickle.load(f)
-            assert status == "ok", f"rank {r} failed:\n{payload}"
-            results.append(payload)
-        return results
+    last_exc = None
+    for attempt in range(2):   # one retry for infra flakes (port races,
+        port = find_free_port()  # spawn hiccups under parallel CI load)
+        with tempfile.TemporaryDirectory() as result_dir:
+            try:
+                mp.spawn(
+                    _worker,
+                    args=(world_size, port, fn, args, kwargs, backend,
+                          result_dir),
+                    nprocs=world_size, join=True, daemon=False)
+            except Exception as e:  # noqa: BLE001
+                # a REAL test failure leaves a pickled traceback; surface it
+                for r in range(world_size):
+                    path = os.path.join(result_dir, f"rank{r}.pkl")
+                    if os.path.exists(path):
+                        with open(path, "rb") as f:
+                            status, payload = pickle.load(f)
+                        if status == "err":
+                            raise AssertionError(
+                                f"rank {r} failed:\n{payload}") from e
+                last_exc = e
+                continue  # infra-level failure: retry once on a new port
+            results = []
+            for r in range(world_size):
+                path = os.path.join(result_dir, f"rank{r}.pkl")
+                assert os.path.exists(path), f"rank {r} produced no result"
+                with open(path, "rb") as f:
+                    status, payload = pickle.load(f)
+                assert status == "ok", f"rank {r} failed:\n{payload}"
+                results.append(payload)
+            return results
+    raise last_exc
