@@ -44,9 +44,16 @@ DEVINL int swzK(int row, int col) {
   return row * D2 + (col ^ ((row & 15) << 3));
 }
 
-// vt_lds [D2][KV]: 128B rows; 16-lane groups read 16 consecutive d-rows
+// vt_lds [D2][KV]: 128B rows.  ds_read_b128 16-lane groups are NOT
+// contiguous on CDNA4 — e.g. {0-3,12-15,20-27} — so a (row&7) xor collides
+// for rows 8 apart within one group (rows 12 vs 20 etc.), which PMC showed
+// as SQ_LDS_BANK_CONFLICT >> SQ_INSTS_MFMA.  slot = (row>>1)^(row>>4) is a
+// bijection per parity class on both real group row-sets {0-3,12-15,20-27}
+// and {4-11,16-19,28-31} (bank = row&1 ? 32:0 | slot*4), making A-frag reads
+// conflict-free; the transpose scatter writes (rows stride 8) hit the 2-way
+// minimum an 8-slot row permits.
 DEVINL int swzV(int row, int col) {
-  return row * KV + (col ^ ((row & 7) << 3));
+  return row * KV + (col ^ ((((row >> 1) ^ (row >> 4)) & 7) << 3));
 }
 
 DEVINL unsigned cvt_pk_bf16(float lo, float hi) {
