@@ -298,8 +298,31 @@ class Bf16ZeroOptimizer:
     def state(self):
         return self.optim.state
 
-    def state_dict(self):
-        return self.optim.state_dict()
+    # Optimizer state is dp-SHARDED: every dp rank must save and reload its
+    # own state_dict (checkpoint.save_checkpoint keys on this attribute and
+    # writes one optim file per dp rank instead of only dp-rank-0's).
+    sharded_state = True
 
+    def state_dict(self):
+        return {
+            "inner": self.optim.state_dict(),
+            "master_flat": self._master_flat.detach().cpu(),
+            "world": self.world,
+        }
+
+    @torch.no_grad()
     def load_state_dict(self, sd):
-        self.optim.load_state_dict(sd)
+        if "master_flat" not in sd:  # plain inner-optimizer dict
+            self.optim.load_state_dict(sd)
+            return
+        if sd.get("world", self.world) != self.world:
+            raise ValueError(
+                f"ZeRO checkpoint was saved at dp world {sd.get('world')}, "
+                f"cannot load at dp world {self.world} (resharding is not "
+                f"supported; keep the dp size or save a dense checkpoint)")
+        self._master_flat.copy_(sd["master_flat"].to(
+            device=self._master_flat.device, dtype=self._master_flat.dtype))
+        self.optim.load_state_dict(sd["inner"])
+        # push the restored (higher-precision) masters back into the model
+        # params on all ranks so params and masters agree bit-for-bit
+        self._sync_params()

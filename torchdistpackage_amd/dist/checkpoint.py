@@ -54,19 +54,38 @@ def _is_save_rank() -> bool:
     return dist.get_rank() == 0
 
 
+def _dp_rank() -> int:
+    if not dist.is_initialized():
+        return 0
+    if tpc.is_mode_inited("data"):
+        return tpc.get_dp_rank()
+    return dist.get_rank()
+
+
 def save_checkpoint(directory: str, step: int, model: torch.nn.Module,
                     optimizer=None, ema=None, scaler=None,
                     extra: Optional[Dict[str, Any]] = None,
                     save_rng: bool = True):
     """Write ``{directory}/ckpt_step{step}{mp_suffix}.pth`` from each model
-    shard's dp-rank-0; rank 0 also writes a ``latest`` pointer file."""
+    shard's dp-rank-0; rank 0 also writes a ``latest`` pointer file.
+
+    Optimizers that mark themselves ``sharded_state = True`` (ZeRO: each dp
+    rank owns a distinct master/optimizer shard) are written as one
+    ``optim_step{N}{mp_suffix}_dp{r}.pth`` file PER dp rank instead of being
+    embedded in dp-rank-0's payload."""
     os.makedirs(directory, exist_ok=True)
+    sharded_opt = optimizer is not None and \
+        getattr(optimizer, "sharded_state", False)
+    if sharded_opt:
+        oname = f"optim_step{step}{get_mp_ckpt_suffix()}_dp{_dp_rank()}.pth"
+        torch.save({"optimizer": optimizer.state_dict()},
+                   os.path.join(directory, oname))
     if _is_save_rank():
         payload: Dict[str, Any] = {
             "step": step,
             "model": model.state_dict(),
         }
-        if optimizer is not None:
+        if optimizer is not None and not sharded_opt:
             payload["optimizer"] = optimizer.state_dict()
         if scaler is not None:
             payload["scaler"] = scaler.state_dict()
@@ -111,7 +130,12 @@ def load_checkpoint(directory: str, model: torch.nn.Module,
     payload = torch.load(os.path.join(directory, name),
                          map_location=map_location, weights_only=False)
     model.load_state_dict(payload["model"], strict=strict)
-    if optimizer is not None and "optimizer" in payload:
+    if optimizer is not None and getattr(optimizer, "sharded_state", False):
+        oname = f"optim_step{step}{get_mp_ckpt_suffix()}_dp{_dp_rank()}.pth"
+        osd = torch.load(os.path.join(directory, oname),
+                         map_location=map_location, weights_only=False)
+        optimizer.load_state_dict(osd["optimizer"])
+    elif optimizer is not None and "optimizer" in payload:
         optimizer.load_state_dict(payload["optimizer"])
     if scaler is not None and "scaler" in payload:
         scaler.load_state_dict(payload["scaler"])
