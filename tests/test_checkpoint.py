@@ -59,8 +59,12 @@ def _ckpt_dp2(rank, world_size, tmpdir):
     files = sorted(os.listdir(tmpdir))
     assert "ckpt_step1.pth" in files and "latest" in files
     model2 = nn.Linear(8, 8)
-    load_checkpoint(tmpdir, model2)
+    ema2 = ShardedEMA(model2, decay=0.9)
+    load_checkpoint(tmpdir, model2, ema=ema2)
     assert torch.equal(model2.weight, model.weight)
+    # EMA shard round-trips through the dense ema file
+    for i in ema._my_idx:
+        assert torch.allclose(ema2._views[i], ema._views[i], atol=1e-7)
     return True
 
 

@@ -118,7 +118,8 @@ def latest_step(directory: str) -> Optional[int]:
 
 
 def load_checkpoint(directory: str, model: torch.nn.Module,
-                    optimizer=None, scaler=None, step: Optional[int] = None,
+                    optimizer=None, scaler=None, ema=None,
+                    step: Optional[int] = None,
                     map_location="cpu", load_rng: bool = True,
                     strict: bool = True) -> Dict[str, Any]:
     """Load this rank's MP shard of the checkpoint; returns the payload."""
@@ -139,6 +140,11 @@ def load_checkpoint(directory: str, model: torch.nn.Module,
         optimizer.load_state_dict(payload["optimizer"])
     if scaler is not None and "scaler" in payload:
         scaler.load_state_dict(payload["scaler"])
+    if ema is not None:
+        ename = f"ema_step{step}{get_mp_ckpt_suffix()}.pth"
+        full = torch.load(os.path.join(directory, ename),
+                          map_location=map_location, weights_only=False)
+        ema.load_state_dict(full)
     if load_rng and "rng" in payload:
         _load_rng_state(payload["rng"])
     if dist.is_initialized():

@@ -117,6 +117,18 @@ class ShardedEMA:
         return out
 
     @torch.no_grad()
+    def load_state_dict(self, full: Dict[str, torch.Tensor]):
+        """Restore from a FULL (dense) EMA state dict, e.g. the
+        ``ema_step{N}.pth`` file written by ``save_checkpoint``; each rank
+        copies only the params of its own shard."""
+        for i in self._my_idx:
+            name = self._names[i]
+            if name not in full:
+                raise KeyError(f"EMA checkpoint is missing param '{name}'")
+            self._views[i].copy_(full[name].detach().reshape(-1).to(
+                device=self._views[i].device, dtype=torch.float32))
+
+    @torch.no_grad()
     def verify_with_gt(self, gt: Dict[str, torch.Tensor],
                        rtol: float = 1e-6, atol: float = 1e-6) -> bool:
         """Compare this rank's shard against a ground-truth dense EMA dict."""
