@@ -26,3 +26,19 @@ print(f"attn fwd us: {t(lambda: ops.ext('x').attn_fwd(q,k,v,o,True,0.088)):.1f}"
 _, lse = ops.ext('x').attn_fwd(q,k,v,o,True,0.088)
 do = torch.randn_like(q); dq=torch.empty_like(q); dk=torch.empty_like(q); dv=torch.empty_like(q)
 print(f"attn bwd us: {t(lambda: ops.ext('x').attn_bwd(do,q,k,v,o,lse,dq,dk,dv,True,0.088)):.1f}")
+
+# bias_gelu at bench fc1 shape (16384, 8192)
+xg = torch.randn(16384, 8192, dtype=torch.bfloat16, device=dev)
+bg = torch.randn(8192, dtype=torch.bfloat16, device=dev)
+dyg = torch.randn_like(xg)
+print(f"bias_gelu fwd us: {t(lambda: ops.ext('x').bias_gelu_fwd(xg, bg)):.1f}")
+print(f"bias_gelu bwd us: {t(lambda: ops.ext('x').bias_gelu_bwd(dyg, xg, bg)):.1f}")
+
+# fused AdamW at 1.3B-param scale (bf16 params + fp32 master, bf16 grads)
+from torchdistpackage_amd.ops.optim import FusedAdamW
+pa = torch.nn.Parameter(torch.zeros(1_300_000_000, dtype=torch.bfloat16,
+                                    device=dev))
+pa.grad = torch.randn_like(pa.data)
+oa = FusedAdamW([pa], lr=1e-3)
+oa.step()
+print(f"adamw 1.3e9 us: {t(lambda: oa.step(), n=10):.1f}")

@@ -7,7 +7,7 @@
 //   - l2norm_sq + inplace scale (grad clip, reference clip_grad_parallel.py)
 //
 // All memory-bound: grid-stride loops, 256-thread blocks, bf16 vectorized as
-// ushort4 (8 B/lane) where applicable, f32 math.
+// ushort4/bf16x8 (8-16 B per lane) where applicable, f32 math.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -19,14 +19,24 @@ constexpr int BLOCK = 256;
 constexpr float GELU_C = 0.7978845608028654f;   // sqrt(2/pi)
 constexpr float GELU_A = 0.044715f;
 
+// tanh via the hardware exp2 unit: tanh(x) = 1 - 2/(exp2(x*2*log2e)+1).
+// Overflow saturates the right way (exp2->inf -> rcp->0 -> +1; exp2->0 -> -1)
+// so no range branches are needed; ulp-exact at bf16 output precision and
+// ~5x fewer instructions than libm tanhf (which rate-limited the fused
+// bias+GELU kernels below ~3.3 TB/s; HBM roofline is ~6).
+DEVINL float tanh_fast(float x) {
+  float e = __builtin_amdgcn_exp2f(x * 2.885390082f);  // 2*log2(e)
+  return 1.f - 2.f / (e + 1.f);
+}
+
 DEVINL float gelu_f(float u) {
-  float t = tanhf(GELU_C * (u + GELU_A * u * u * u));
+  float t = tanh_fast(GELU_C * (u + GELU_A * u * u * u));
   return 0.5f * u * (1.f + t);
 }
 
 DEVINL float gelu_df(float u) {
   float u2 = u * u;
-  float t = tanhf(GELU_C * (u + GELU_A * u * u2));
+  float t = tanh_fast(GELU_C * (u + GELU_A * u * u2));
   return 0.5f * (1.f + t) +
          0.5f * u * (1.f - t * t) * GELU_C * (1.f + 3.f * GELU_A * u2);
 }
@@ -37,23 +47,24 @@ __global__ void bias_gelu_fwd_bf16(const unsigned short* __restrict__ x,
                                    const unsigned short* __restrict__ bias,
                                    unsigned short* __restrict__ y,
                                    long n, int D, int has_bias) {
-  long i0 = ((long)blockIdx.x * BLOCK + threadIdx.x) * 4;
-  long stride = (long)gridDim.x * BLOCK * 4;
+  long i0 = ((long)blockIdx.x * BLOCK + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * BLOCK * 8;
   for (long i = i0; i < n; i += stride) {
-    if (i + 4 <= n) {
-      ushort4 xv = *(const ushort4*)(x + i);
-      int c = (int)(i % D);
-      float b0 = 0, b1 = 0, b2 = 0, b3 = 0;
-      if (has_bias) {  // D % 4 == 0 guaranteed by caller for the vector path
-        ushort4 bv = *(const ushort4*)(bias + c);
-        b0 = bf2f(bv.x); b1 = bf2f(bv.y); b2 = bf2f(bv.z); b3 = bf2f(bv.w);
+    if (i + 8 <= n) {
+      bf16x8 xv = *(const bf16x8*)(x + i);
+      // D % 8 == 0 guaranteed by caller for the vector path
+      bf16x8 bv;
+      if (has_bias) bv = *(const bf16x8*)(bias + (int)(i % D));
+      const unsigned short* xs = (const unsigned short*)&xv;
+      const unsigned short* bs = (const unsigned short*)&bv;
+      bf16x8 out;
+      unsigned short* os = (unsigned short*)&out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float b = has_bias ? bf2f(bs[j]) : 0.f;
+        os[j] = f2bf(gelu_f(bf2f(xs[j]) + b));
       }
-      ushort4 out;
-      out.x = f2bf(gelu_f(bf2f(xv.x) + b0));
-      out.y = f2bf(gelu_f(bf2f(xv.y) + b1));
-      out.z = f2bf(gelu_f(bf2f(xv.z) + b2));
-      out.w = f2bf(gelu_f(bf2f(xv.w) + b3));
-      *(ushort4*)(y + i) = out;
+      *(bf16x8*)(y + i) = out;
     } else {
       for (long j = i; j < n; ++j) {
         float b = has_bias ? bf2f(bias[j % D]) : 0.f;
@@ -68,24 +79,25 @@ __global__ void bias_gelu_bwd_bf16(const unsigned short* __restrict__ dy,
                                    const unsigned short* __restrict__ bias,
                                    unsigned short* __restrict__ dx,
                                    long n, int D, int has_bias) {
-  long i0 = ((long)blockIdx.x * BLOCK + threadIdx.x) * 4;
-  long stride = (long)gridDim.x * BLOCK * 4;
+  long i0 = ((long)blockIdx.x * BLOCK + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * BLOCK * 8;
   for (long i = i0; i < n; i += stride) {
-    if (i + 4 <= n) {
-      ushort4 xv = *(const ushort4*)(x + i);
-      ushort4 gv = *(const ushort4*)(dy + i);
-      int c = (int)(i % D);
-      float b0 = 0, b1 = 0, b2 = 0, b3 = 0;
-      if (has_bias) {
-        ushort4 bv = *(const ushort4*)(bias + c);
-        b0 = bf2f(bv.x); b1 = bf2f(bv.y); b2 = bf2f(bv.z); b3 = bf2f(bv.w);
+    if (i + 8 <= n) {
+      bf16x8 xv = *(const bf16x8*)(x + i);
+      bf16x8 gv = *(const bf16x8*)(dy + i);
+      bf16x8 bv;
+      if (has_bias) bv = *(const bf16x8*)(bias + (int)(i % D));
+      const unsigned short* xs = (const unsigned short*)&xv;
+      const unsigned short* gs = (const unsigned short*)&gv;
+      const unsigned short* bs = (const unsigned short*)&bv;
+      bf16x8 out;
+      unsigned short* os = (unsigned short*)&out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float b = has_bias ? bf2f(bs[j]) : 0.f;
+        os[j] = f2bf(bf2f(gs[j]) * gelu_df(bf2f(xs[j]) + b));
       }
-      ushort4 out;
-      out.x = f2bf(bf2f(gv.x) * gelu_df(bf2f(xv.x) + b0));
-      out.y = f2bf(bf2f(gv.y) * gelu_df(bf2f(xv.y) + b1));
-      out.z = f2bf(bf2f(gv.z) * gelu_df(bf2f(xv.z) + b2));
-      out.w = f2bf(bf2f(gv.w) * gelu_df(bf2f(xv.w) + b3));
-      *(ushort4*)(dx + i) = out;
+      *(bf16x8*)(dx + i) = out;
     } else {
       for (long j = i; j < n; ++j) {
         float b = has_bias ? bf2f(bias[j % D]) : 0.f;
@@ -199,8 +211,8 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor bias) {
   int has_bias = bias.numel() > 0 ? 1 : 0;
   auto stream = at::cuda::getCurrentHIPStream();
   if (x.scalar_type() == torch::kBFloat16) {
-    TORCH_CHECK(!has_bias || (D % 4 == 0), "D must be multiple of 4");
-    hipLaunchKernelGGL(bias_gelu_fwd_bf16, dim3(ew_grid(n, 4)), dim3(BLOCK), 0,
+    TORCH_CHECK(!has_bias || (D % 8 == 0), "D must be multiple of 8");
+    hipLaunchKernelGGL(bias_gelu_fwd_bf16, dim3(ew_grid(n, 8)), dim3(BLOCK), 0,
                        stream, (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)bias.data_ptr(),
                        (unsigned short*)y.data_ptr(), n, D, has_bias);
@@ -221,7 +233,8 @@ torch::Tensor bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
   int has_bias = bias.numel() > 0 ? 1 : 0;
   auto stream = at::cuda::getCurrentHIPStream();
   if (x.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(ew_grid(n, 4)), dim3(BLOCK), 0,
+    TORCH_CHECK(!has_bias || (D % 8 == 0), "D must be multiple of 8");
+    hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(ew_grid(n, 8)), dim3(BLOCK), 0,
                        stream, (const unsigned short*)dy.data_ptr(),
                        (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)bias.data_ptr(),
@@ -327,6 +340,67 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
   float* master = (float*)mptrs[pid];  // per-param master (MAY alias param)
   void* pp = (void*)pptrs[pid];
   const void* gp = (const void*)gptrs[pid];
+  // HBM-bound: ~26 B/element of read+write traffic, so dwordx4 accesses are
+  // what decides the rate.  All pointers/offsets are 16B-aligned in practice
+  // (torch allocations + 64Ki chunking); fall back to scalars otherwise.
+  const bool vec4 = ((off | mvbase) & 3) == 0 &&
+      (((long)master | (long)pp | (long)gp) & 15) == 0;
+  if (vec4) {
+    const long nv = (end - off) >> 2;   // full float4 groups in this chunk
+    for (long g = threadIdx.x; g < nv; g += blockDim.x) {
+      const long i = off + g * 4;
+      const long k = mvbase + i;
+      float4 gi4;
+      if (gp == nullptr) gi4 = make_float4(0.f, 0.f, 0.f, 0.f);
+      else if (grad_bf16) {
+        ushort4 gu = *(const ushort4*)((const unsigned short*)gp + i);
+        gi4 = make_float4(bf2f(gu.x), bf2f(gu.y), bf2f(gu.z), bf2f(gu.w));
+      } else {
+        gi4 = *(const float4*)((const float*)gp + i);
+      }
+      float4 p4 = *(const float4*)(master + i);
+      float4 m4 = *(const float4*)(m + k);
+      float4 v4 = *(const float4*)(v + k);
+      float* pi = (float*)&p4;
+      float* mi = (float*)&m4;
+      float* vi = (float*)&v4;
+      const float* gi = (const float*)&gi4;
+      ushort4 pb;
+      unsigned short* pbs = (unsigned short*)&pb;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float pp_ = pi[j] * (1.f - lr * wd);
+        float mm_ = mi[j] * beta1 + gi[j] * (1.f - beta1);
+        float vv_ = vi[j] * beta2 + gi[j] * gi[j] * (1.f - beta2);
+        mi[j] = mm_;
+        vi[j] = vv_;
+        pp_ = pp_ - lr / bc1 * mm_ / (sqrtf(vv_ / bc2) + eps);
+        pi[j] = pp_;
+        pbs[j] = f2bf(pp_);
+      }
+      *(float4*)(m + k) = m4;
+      *(float4*)(v + k) = v4;
+      *(float4*)(master + i) = p4;
+      if (param_bf16) *(ushort4*)((unsigned short*)pp + i) = pb;
+    }
+    // scalar tail: at most 3 elements at the chunk end
+    for (long i = off + nv * 4 + threadIdx.x; i < end; i += blockDim.x) {
+      float gi;
+      if (gp == nullptr) gi = 0.f;
+      else if (grad_bf16) gi = bf2f(((const unsigned short*)gp)[i]);
+      else gi = ((const float*)gp)[i];
+      const long k = mvbase + i;
+      float pi = master[i] * (1.f - lr * wd);
+      float mi = m[k] * beta1 + gi * (1.f - beta1);
+      float vi = v[k] * beta2 + gi * gi * (1.f - beta2);
+      m[k] = mi;
+      v[k] = vi;
+      pi = pi - lr / bc1 * mi / (sqrtf(vi / bc2) + eps);
+      master[i] = pi;
+      if (param_bf16) ((unsigned short*)pp)[i] = f2bf(pi);
+    }
+    return;
+  }
   for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
     float gi;
     if (gp == nullptr) gi = 0.f;
