@@ -374,3 +374,4 @@ def test_graphed_step():
     torch.cuda.synchronize()
     assert torch.isfinite(loss_box["loss"])
     assert not torch.equal(p0, model[0].weight), "params must update"
+

@@ -148,3 +148,4 @@ def test_rope_rotation_property():
     d1 = torch.dot(rot(q, 3), rot(k, 5))
     d2 = torch.dot(rot(q, 10), rot(k, 12))
     assert torch.allclose(d1, d2, atol=1e-4)
+
