@@ -616,6 +616,10 @@ void attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 void attn_bwd_dq_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
                     torch::Tensor dq, bool causal, double scale);
+void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                     torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                     bool causal, double scale);
 void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                       torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
                       torch::Tensor dk, torch::Tensor dv, bool causal,
@@ -693,8 +697,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
 
   static const bool force_v1b = std::getenv("TDPA_ATTN_V1") != nullptr;
   if (D == 128 && !force_v1b) {
-    attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
-    attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
+    attn_bwd_v2_all(dout, q, k, v, lse, delta, dq, dk, dv, causal, scale);
     HIP_CHECK_LAST();
     return {dq, dk, dv};
   }

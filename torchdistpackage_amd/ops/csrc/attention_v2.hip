@@ -574,7 +574,11 @@ void attn_bwd_dq_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
 
 namespace {
 
-template <bool CAUSAL, bool DK_PASS>
+// WRITE_DS (dk pass only): also store dS bf16 to ds_out[bh][q][key-padded],
+// coalesced (each 32-lane half writes one q-row, 32 consecutive keys).  The
+// dq-lite kernel then consumes dS directly instead of recomputing
+// S/P/dP per tile (drops 32 of dq's 48 MFMAs plus its V/dO staging).
+template <bool CAUSAL, bool DK_PASS, bool WRITE_DS>
 __launch_bounds__(512)
 __global__ void attn_bwd_dkdv_v2_kernel(
     const unsigned short* __restrict__ q,
@@ -584,6 +588,7 @@ __global__ void attn_bwd_dkdv_v2_kernel(
     const float* __restrict__ lse,
     const float* __restrict__ delta,
     unsigned short* __restrict__ out,   // dk (DK_PASS) or dv
+    unsigned short* __restrict__ ds_out, long spad,
     Strides2 qs, Strides2 ks, Strides2 vs, Strides2 dos, Strides2 outs,
     int B, int H, int S, float scale, int q_per_kv) {
   // LDS: Q and dO tiles (A-operands), plus the transposed tile the pass's
@@ -701,6 +706,8 @@ __global__ void attn_bwd_dkdv_v2_kernel(
             __expf(s_acc[r] * scale - lse_lds[buf][qrel]) : 0.f;
         s_acc[r] = DK_PASS
             ? p * (dp_acc[r] - del_lds[buf][qrel]) * scale : p;
+        if (DK_PASS && WRITE_DS && qrow < S)
+          ds_out[((long)bh * S + qrow) * spad + key] = f2bf(s_acc[r]);
       }
 
       // B-fragments over the q dimension (exchange as in the forward)
@@ -774,28 +781,178 @@ void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   };
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
-#define L_KV2(CC)                                                             \
+#define L_KV2(CC, WDS, DSP, SPAD)                                             \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false>), grid, block, 0,  \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false, false>), grid,     \
+                       block, 0,                                              \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
                        (const unsigned short*)v.data_ptr(),                   \
                        (const unsigned short*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (unsigned short*)dv.data_ptr(), get(q), get(k),        \
-                       get(v), get(dout), get(dv), B, H, S, (float)scale,     \
-                       q_per_kv);                                             \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true>), grid, block, 0,   \
+                       (unsigned short*)dv.data_ptr(), nullptr, 0, get(q),    \
+                       get(k), get(v), get(dout), get(dv), B, H, S,           \
+                       (float)scale, q_per_kv);                               \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true, WDS>), grid,        \
+                       block, 0,                                              \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
                        (const unsigned short*)v.data_ptr(),                   \
                        (const unsigned short*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (unsigned short*)dk.data_ptr(), get(q), get(k),        \
-                       get(v), get(dout), get(dk), B, H, S, (float)scale,     \
-                       q_per_kv);                                             \
+                       (unsigned short*)dk.data_ptr(), DSP, SPAD, get(q),     \
+                       get(k), get(v), get(dout), get(dk), B, H, S,           \
+                       (float)scale, q_per_kv);                               \
   } while (0)
-  if (causal) L_KV2(true); else L_KV2(false);
+  if (causal) L_KV2(true, false, nullptr, 0);
+  else L_KV2(false, false, nullptr, 0);
+  HIP_CHECK_LAST();
+}
+
+// ===========================================================================
+// dq-lite: consumes the dS tile the dk pass stored instead of recomputing
+// S, P and dP per key-tile (48 -> 16 MFMAs per tile; no V/dO staging, no
+// exp, no lse/delta).  dS fragments load straight from global: the
+// [bh][q][key] layout makes each B-fragment one 16-byte read per lane whose
+// element order IS the MFMA k-order (key = kt*32 + kc*16 + hi*8 + j).
+// ===========================================================================
+
+namespace {
+
+template <bool CAUSAL>
+__launch_bounds__(512)
+__global__ void attn_bwd_dq_lite_kernel(
+    const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ dsb,  // [bh][S][spad] bf16
+    unsigned short* __restrict__ dq,
+    Strides2 ks, Strides2 dqs,
+    int B, int H, int S, long spad, int q_per_kv) {
+  __shared__ __attribute__((aligned(16))) unsigned short kt_lds[2][D2 * KV];
+
+  const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
+  const int hkv = hh / q_per_kv;
+  const int qtile = CAUSAL ? (gridDim.x - 1 - blockIdx.x) : blockIdx.x;
+  const int qbase = qtile * QT2;
+  if (qbase >= S) return;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const int qrow = qbase + wid * QW + l31;
+  const unsigned short* dsp = dsb +
+      ((long)bh * S + (qrow < S ? qrow : S - 1)) * spad;
+
+  f32x16 acc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) acc[ds] = (f32x16)(0.f);
+
+  const int kv_end = CAUSAL ? min(S, qbase + QT2) : S;
+
+  auto stage_kt = [&](int kt0, int b) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int key = idx / D2;
+      int col = idx % D2;
+      int gkey = kt0 + key;
+      bf16x8_v kv8 = gkey < S ? pack8v(kp + (long)gkey * ks.s + col)
+                              : (bf16x8_v)(__bf16)0.f;
+      const unsigned short* ksrc = (const unsigned short*)&kv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        kt_lds[b][swzV(col + j, key)] = ksrc[j];
+    }
+  };
+  stage_kt(0, 0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
+    bf16x8_v db[2][2];
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+        db[kt][kc] = pack8v(dsp + kt0 + kt * 32 + kc * 16 + hi * 8);
+
+    // dq^T += K^T @ dS^T
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8_v a_kt = pack8v(&kt_lds[buf][swzV(ds * 32 + l31,
+                                                   kt * 32 + kc * 16 +
+                                                   hi * 8)]);
+          acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_kt, db[kt][kc], acc[ds], 0, 0, 0);
+        }
+      }
+    }
+
+    if (kt0 + KV < kv_end) stage_kt(kt0 + KV, buf ^ 1);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  if (qrow < S) {
+    unsigned short* dqp = dq + bb * dqs.b + hh * dqs.h + (long)qrow * dqs.s;
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int blk = 0; blk < 4; ++blk) {
+        int d0 = ds * 32 + 8 * blk + 4 * hi;
+        unsigned short out4[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          out4[j] = f2bf(acc[ds][4 * blk + j]);
+        *(ushort4*)(dqp + d0) = *(ushort4*)out4;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// Combined backward: dv pass, dk pass storing dS, dq-lite consuming it.
+// TDPA_DQ_RECOMPUTE falls back to the independent recomputing dq kernel
+// (and no dS workspace) for A/B and as an escape hatch.
+void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                     torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                     bool causal, double scale) {
+  static const bool recompute = std::getenv("TDPA_DQ_RECOMPUTE") != nullptr;
+  if (recompute) {
+    attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
+    attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
+    return;
+  }
+  const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  const int q_per_kv = H / Hkv;
+  TORCH_CHECK(D == 128);
+  auto get = [](const torch::Tensor& t) {
+    return Strides2{t.stride(0), t.stride(1), t.stride(2)};
+  };
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
+  const long spad = (long)grid.x * QT2;
+  auto dsw = torch::empty({(long)B * H, (long)S, spad}, q.options());
+  unsigned short* dsp = (unsigned short*)dsw.data_ptr();
+  if (causal) L_KV2(true, true, dsp, spad);
+  else L_KV2(false, true, dsp, spad);
 #undef L_KV2
+#define L_DQL(CC)                                                             \
+  hipLaunchKernelGGL((attn_bwd_dq_lite_kernel<CC>), grid, block, 0, stream,   \
+                     (const unsigned short*)k.data_ptr(), dsp,                \
+                     (unsigned short*)dq.data_ptr(), get(k), get(dq),         \
+                     B, H, S, spad, q_per_kv)
+  if (causal) L_DQL(true); else L_DQL(false);
+#undef L_DQL
   HIP_CHECK_LAST();
 }
