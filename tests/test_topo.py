@@ -76,3 +76,25 @@ def _ring_check(rank, world_size):
 
 def test_pipe_ring_world2():
     run_distributed(_ring_check, world_size=2)
+
+
+def _first_last_helpers(rank, world_size):
+    from torchdistpackage_amd.dist.topo import tpc
+    tpc.setup_process_groups([("data", 2), ("tensor", 2)])
+    # first group of 'tensor' = ranks [0, 1]; of 'data' = [0, 2]
+    assert tpc.is_first_group("tensor") == (rank in (0, 1))
+    assert tpc.is_first_group("data") == (rank in (0, 2))
+    assert tpc.is_first_in_group("tensor") == (rank % 2 == 0)
+    assert tpc.is_last_in_group("tensor") == (rank % 2 == 1)
+    assert tpc.is_last_in_data_group() == (rank >= 2)
+    assert tpc.is_last_in_tensor_group() == (rank % 2 == 1)
+    assert tpc.all_ranks() == [0, 1, 2, 3]
+    assert tpc.all_ranks("tensor") == [[0, 1], [2, 3]]
+    assert tpc.all_dp_ranks() == [[0, 2], [1, 3]]
+    # model = transpose of data
+    assert tpc.is_first_in_model_group() == (tpc.get_mp_rank() == 0)
+    return True
+
+
+def test_first_last_helpers():
+    run_distributed(_first_last_helpers, world_size=4)

@@ -88,6 +88,7 @@ class ProcessTopology(metaclass=_Singleton):
     def _reset(self):
         self._groups: Dict[str, dist.ProcessGroup] = {}
         self._ranks_in_group: Dict[str, List[int]] = {}
+        self._ranks_all: Dict[str, List[List[int]]] = {}
         self._group_rank: Dict[str, int] = {}
         self._group_size: Dict[str, int] = {}
         self._axis_order: List[str] = []
@@ -163,6 +164,7 @@ class ProcessTopology(metaclass=_Singleton):
         # Axis present globally even if this rank's group wasn't matched (can't
         # happen — rank lists cover all ranks — but keep the invariant checked).
         assert name in self._groups, f"rank {rank} not covered by axis {name}"
+        self._ranks_all[name] = [list(r) for r in rank_lists]
         self._axis_sizes.setdefault(name, size)
 
     def build_moe_groups(self, moe_dp_size: int, moe_ep_size: int):
@@ -271,13 +273,42 @@ class ProcessTopology(metaclass=_Singleton):
         i = self._group_rank[mode]
         return ranks[(i + 1) % len(ranks)]
 
-    def all_ranks(self) -> List[int]:
-        return list(range(dist.get_world_size()))
+    def all_ranks(self, mode: Optional[str] = None):
+        """No mode: all global ranks.  With a mode: every group's rank list
+        for that axis (reference all_ranks, process_topo.py:242-246)."""
+        if mode is None:
+            return list(range(dist.get_world_size()))
+        if mode not in self._ranks_all:
+            raise RuntimeError(f"{mode} is not initialized")
+        return self._ranks_all[mode]
+
+    def all_dp_ranks(self) -> List[List[int]]:
+        return self.all_ranks("data")
+
+    def is_first_in_group(self, mode: str) -> bool:
+        return self._group_rank.get(mode, 0) == 0
+
+    def is_last_in_group(self, mode: str) -> bool:
+        return self._group_rank.get(mode, 0) == self._group_size.get(mode, 1) - 1
+
+    def is_last_in_data_group(self) -> bool:
+        return self.is_last_in_group("data")
+
+    def is_last_in_tensor_group(self) -> bool:
+        return self.is_last_in_group("tensor")
+
+    def is_first_in_model_group(self) -> bool:
+        return self.is_first_in_group("model")
+
+    def is_last_in_model_group(self) -> bool:
+        return self.is_last_in_group("model")
 
     def is_first_group(self, mode: str) -> bool:
-        """True if this rank's mode-group is the lexicographically first one."""
-        return dist.get_rank() == min(self._ranks_in_group[mode]) or \
-            min(self._ranks_in_group[mode]) == 0
+        """True if this rank's mode-group IS the first group of that axis
+        (reference process_topo.py:248-259)."""
+        if mode not in self._ranks_all:
+            raise RuntimeError(f"{mode} is not initialized")
+        return self._ranks_in_group[mode] == self._ranks_all[mode][0]
 
     @property
     def inited(self) -> bool:
