@@ -400,3 +400,14 @@ def allreduce_sequence_parallel_grads(module: "nn.Module"):
     for g in grads:
         g.copy_(flat[off:off + g.numel()].view_as(g))
         off += g.numel()
+
+
+# --------------------------------------------------------------------------
+# Migration aliases for users of the reference package (tp_utils.py there
+# exposes these names; ``is_squence_parallel_tensor`` keeps the reference's
+# spelling so ported code imports unchanged).
+# --------------------------------------------------------------------------
+
+get_tensor_model_parallel_world_size = get_tp_size
+maybe_gather_from_sequence_parallel = maybe_gather_for_sequence_parallel
+is_squence_parallel_tensor = is_sequence_parallel
