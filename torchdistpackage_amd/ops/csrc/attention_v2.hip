@@ -639,6 +639,30 @@ __global__ void attn_bwd_dkdv_v2_kernel(
 
   const int qt_start = CAUSAL ? kbase : 0;
 
+  // T14 register staging for the dv pass: global loads for tile t+1 issue
+  // right after tile t+1's LDS write, so HBM latency crosses a whole tile
+  // of MFMAs (the forward's scheme; measured on it at +?% — see notes).
+  // The dk pass sits at 248 VGPRs and keeps the direct stage instead.
+  // T14 register staging for the dv pass only: the dk pass sits at 248
+  // VGPRs and prefetch regs push it into (measured net-negative) spills.
+  bf16x8_v q_reg[DK_PASS ? 1 : 2], do_reg[DK_PASS ? 1 : 2];
+  auto load_regs = [&](int qt0) {
+    if (DK_PASS) return;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int row = idx / D2;
+      int col = idx % D2;
+      int grow = qt0 + row;
+      if (grow < S) {
+        q_reg[DK_PASS ? 0 : c] = pack8v(qp + (long)grow * qs.s + col);
+        do_reg[DK_PASS ? 0 : c] = pack8v(dop + (long)grow * dos.s + col);
+      } else {
+        q_reg[DK_PASS ? 0 : c] = (bf16x8_v)(__bf16)0.f;
+        do_reg[DK_PASS ? 0 : c] = (bf16x8_v)(__bf16)0.f;
+      }
+    }
+  };
   auto stage_kv = [&](int qt0, int b) {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
@@ -647,12 +671,17 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       int col = idx % D2;
       int grow = qt0 + row;
       bf16x8_v qv, dv8;
-      if (grow < S) {
-        qv = pack8v(qp + (long)grow * qs.s + col);
-        dv8 = pack8v(dop + (long)grow * dos.s + col);
+      if (DK_PASS) {
+        if (grow < S) {
+          qv = pack8v(qp + (long)grow * qs.s + col);
+          dv8 = pack8v(dop + (long)grow * dos.s + col);
+        } else {
+          qv = (bf16x8_v)(__bf16)0.f;
+          dv8 = (bf16x8_v)(__bf16)0.f;
+        }
       } else {
-        qv = (bf16x8_v)(__bf16)0.f;
-        dv8 = (bf16x8_v)(__bf16)0.f;
+        qv = q_reg[DK_PASS ? 0 : c];
+        dv8 = do_reg[DK_PASS ? 0 : c];
       }
       *(bf16x8_v*)&q_lds[b][swzK(row, col)] = qv;
       if (DK_PASS)
@@ -671,7 +700,9 @@ __global__ void attn_bwd_dkdv_v2_kernel(
         del_lds[b][tid] = grow < S ? delp[grow] : 0.f;
     }
   };
+  load_regs(qt_start);
   stage_kv(qt_start, 0);
+  if (qt_start + KV < S) load_regs(qt_start + KV);
   __syncthreads();
 
   int buf = 0;
@@ -743,7 +774,10 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       }
     }
 
-    if (qt0 + KV < S) stage_kv(qt0 + KV, buf ^ 1);
+    if (qt0 + KV < S) {
+      stage_kv(qt0 + KV, buf ^ 1);
+      if (qt0 + 2 * KV < S) load_regs(qt0 + 2 * KV);
+    }
     __syncthreads();
     buf ^= 1;
   }
@@ -852,22 +886,31 @@ __global__ void attn_bwd_dq_lite_kernel(
 
   const int kv_end = CAUSAL ? min(S, qbase + QT2) : S;
 
-  auto stage_kt = [&](int kt0, int b) {
+  bf16x8_v k_reg[2];
+  auto load_regs = [&](int kt0) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int gkey = kt0 + idx / D2;
+      k_reg[c] = gkey < S ? pack8v(kp + (long)gkey * ks.s + idx % D2)
+                          : (bf16x8_v)(__bf16)0.f;
+    }
+  };
+  auto stage_kt = [&](int b) {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
       int idx = tid * 8 + c * 4096;
       int key = idx / D2;
       int col = idx % D2;
-      int gkey = kt0 + key;
-      bf16x8_v kv8 = gkey < S ? pack8v(kp + (long)gkey * ks.s + col)
-                              : (bf16x8_v)(__bf16)0.f;
-      const unsigned short* ksrc = (const unsigned short*)&kv8;
+      const unsigned short* ksrc = (const unsigned short*)&k_reg[c];
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         kt_lds[b][swzV(col + j, key)] = ksrc[j];
     }
   };
-  stage_kt(0, 0);
+  load_regs(0);
+  stage_kt(0);
+  if (KV < kv_end) load_regs(KV);
   __syncthreads();
 
   int buf = 0;
@@ -895,7 +938,10 @@ __global__ void attn_bwd_dq_lite_kernel(
       }
     }
 
-    if (kt0 + KV < kv_end) stage_kt(kt0 + KV, buf ^ 1);
+    if (kt0 + KV < kv_end) {
+      stage_kt(buf ^ 1);
+      if (kt0 + 2 * KV < kv_end) load_regs(kt0 + 2 * KV);
+    }
     __syncthreads();
     buf ^= 1;
   }
