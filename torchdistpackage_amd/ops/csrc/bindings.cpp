@@ -25,7 +25,6 @@ void multi_adamw_step(torch::Tensor cpid, torch::Tensor coff,
                       double eps, double wd, bool param_bf16, bool grad_bf16);
 void ema_update(torch::Tensor ema, torch::Tensor p, double decay);
 torch::Tensor l2norm_sq(torch::Tensor x);
-torch::Tensor colsum_rows_f32(torch::Tensor partial);
 void scale_inplace(torch::Tensor x, double s);
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
@@ -53,7 +52,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_adamw_step", &multi_adamw_step);
   m.def("ema_update", &ema_update);
   m.def("l2norm_sq", &l2norm_sq);
-  m.def("colsum_rows_f32", &colsum_rows_f32);
   m.def("scale_inplace", &scale_inplace);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);

@@ -180,24 +180,6 @@ __global__ void l2norm_kernel(const T* __restrict__ x, float* __restrict__ out,
   if (threadIdx.x == 0) atomicAdd(out, acc);
 }
 
-// -------- fold fp32 partial rows (nr, D) -> (D) --------------------------
-// One lean launch replacing torch's reduce dispatch for the tiny per-block
-// partial buffers the norm backwards emit.  (A full bf16 column-sum kernel
-// was tried for dbias and REMOVED: torch's bf16 reduce already streams at
-// ~4.9 TB/s on (16384, 6144) — see profiles/r01_notes.md v11.)
-
-__global__ void colsum_stage2(const float* __restrict__ partial,
-                              float* __restrict__ out, long nr, long D) {
-  const long col0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  if (col0 >= D) return;
-  float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
-  for (long r = 0; r < nr; ++r) {
-    float4 v = *(const float4*)(partial + r * D + col0);
-    acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
-  }
-  *(float4*)(out + col0) = acc;
-}
-
 template <typename T>
 __global__ void scale_kernel(T* __restrict__ x, long n, float s) {
   long i0 = (long)blockIdx.x * BLOCK + threadIdx.x;
@@ -323,22 +305,6 @@ void scale_inplace(torch::Tensor x, double s) {
                        stream, x.data_ptr<float>(), n, (float)s);
   }
   HIP_CHECK_LAST();
-}
-
-// fold already-materialized fp32 partial rows (nr, D) -> (D); used by the
-// layernorm/rmsnorm backward wrappers instead of torch .sum(0)
-torch::Tensor colsum_rows_f32(torch::Tensor partial) {
-  TORCH_CHECK(partial.is_cuda() && partial.is_contiguous() &&
-              partial.scalar_type() == torch::kFloat && partial.dim() == 2);
-  const long nr = partial.size(0), D = partial.size(1);
-  TORCH_CHECK(D % 4 == 0);
-  auto out = torch::empty({D}, partial.options());
-  auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid2((unsigned)((D / 4 + BLOCK - 1) / BLOCK));
-  hipLaunchKernelGGL(colsum_stage2, grid2, dim3(BLOCK), 0, stream,
-                     partial.data_ptr<float>(), out.data_ptr<float>(), nr, D);
-  HIP_CHECK_LAST();
-  return out;
 }
 
 // -------- multi-tensor AdamW: one launch for ALL params ---------------------

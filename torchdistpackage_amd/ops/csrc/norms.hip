@@ -597,9 +597,6 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
   return {y, rstd};
 }
 
-// fast fold of fp32 partial rows (defined in elementwise.hip)
-torch::Tensor colsum_rows_f32(torch::Tensor partial);
-
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor rstd) {
   const int D = x.size(-1);
@@ -627,7 +624,7 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
     else if (nc == 3) RMS_REG(3); else RMS_REG(4);
 #undef RMS_REG
     HIP_CHECK_LAST();
-    return {dx, colsum_rows_f32(dw_partial).to(w.scalar_type())};
+    return {dx, dw_partial.sum(0).to(w.scalar_type())};
   } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(rmsnorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
@@ -721,8 +718,8 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
     else if (nc == 3) LN_REG(3); else LN_REG(4);
 #undef LN_REG
     HIP_CHECK_LAST();
-    return {dx, colsum_rows_f32(dw_partial).to(w.scalar_type()),
-            colsum_rows_f32(db_partial).to(w.scalar_type())};
+    return {dx, dw_partial.sum(0).to(w.scalar_type()),
+            db_partial.sum(0).to(w.scalar_type())};
   } else if (x.scalar_type() == torch::kBFloat16 && D % 8 == 0) {
     hipLaunchKernelGGL(layernorm_bwd_bf16v8, grid, block, lds, stream,
                        (const unsigned short*)dy.data_ptr(),
