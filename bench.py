@@ -38,7 +38,7 @@ def parse_args():
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--model", type=str, default="gpt2_1.3b",
                     choices=["gpt2_1.3b", "gpt2_small", "tiny", "llama_8b",
-                             "moe_8x"])
+                             "moe_8x", "moe_tiny"])
     ap.add_argument("--graph", action="store_true",
                     help="capture the whole train step in one hipGraph "
                          "(world_size==1 only)")
@@ -100,6 +100,11 @@ def main():
     elif args.model == "moe_8x":
         from torchdistpackage_amd.models.moe_model import mixtral_style_8x
         cfg = mixtral_style_8x()
+    elif args.model == "moe_tiny":
+        from torchdistpackage_amd.models.moe_model import MoEConfig
+        cfg = MoEConfig(vocab_size=512, n_layer=2, n_head=2, dim=128,
+                        max_seq=args.seq, num_experts=4, top_k=2,
+                        hidden_mult=2)
     else:
         cfg = GPT2Config(vocab_size=2048, n_layer=4, n_head=8, dim=512,
                          max_seq=args.seq)
@@ -161,12 +166,13 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
     if args.model == "llama_8b":
         from torchdistpackage_amd.models.llama import LlamaModel
         model = LlamaModel(cfg, device=dev, dtype=dtype)
-    elif args.model == "moe_8x":
+    elif args.model.startswith("moe"):
         from torchdistpackage_amd.models.moe_model import MoEModel
-        model = MoEModel(cfg, device=dev, dtype=dtype)
         if world > 1:
+            # groups must exist BEFORE the model captures its EP group
             ep = min(dp, cfg.num_experts)
             tpc.build_moe_groups(moe_dp_size=dp // ep, moe_ep_size=ep)
+        model = MoEModel(cfg, device=dev, dtype=dtype)
     else:
         model = GPT2Model(cfg, device=dev, dtype=dtype)
 
@@ -185,7 +191,7 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
     else:
         if world > 1 and dp > 1:
             model = NaiveDdp(model, group=tpc.get_group("data"))
-            if args.model == "moe_8x":
+            if args.model.startswith("moe"):
                 from torchdistpackage_amd.ddp import create_moe_dp_hooks
                 inner = model.module
                 if tpc.get_group_size("moe_dp") > 1:

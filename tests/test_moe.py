@@ -216,3 +216,29 @@ def _moe_full_composition(rank, world_size):
 
 def test_moe_full_composition_world4():
     run_distributed(_moe_full_composition, world_size=4)
+
+
+def _bench_order_regression(rank, world_size):
+    """Regression: bench.py once built the MoE model BEFORE build_moe_groups,
+    so layers captured ep_size=1 with group=None and the dispatch emulation
+    treated None as the world group (IndexError).  Groups-first must work and
+    the single-EP capture must be harmless."""
+    import torch
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.models.moe_model import MoEConfig, MoEModel
+
+    tpc.setup_process_groups([("data", world_size)])
+    tpc.build_moe_groups(moe_dp_size=1, moe_ep_size=world_size)
+    cfg = MoEConfig(vocab_size=64, n_layer=1, n_head=2, dim=32, max_seq=16,
+                    num_experts=world_size * 2, top_k=2, hidden_mult=2)
+    m = MoEModel(cfg)
+    x = torch.randint(0, 64, (2, 16))
+    out = m(x, labels=x)
+    out["loss"].backward()
+    assert torch.isfinite(out["loss"])
+
+    return True
+
+
+def test_bench_order_regression():
+    run_distributed(_bench_order_regression, world_size=2)

@@ -208,8 +208,11 @@ class ExpertParallelMoE(nn.Module):
             out_splits = [int(counts.sum())]
 
         dispatched = xt[token_of]                                # (N*k, D)
-        received = _AllToAll.apply(dispatched, in_splits, out_splits,
-                                   self.ep_group)
+        if self.ep_size > 1:
+            received = _AllToAll.apply(dispatched, in_splits, out_splits,
+                                       self.ep_group)
+        else:
+            received = dispatched   # single EP rank: no exchange
 
         # received tokens are ordered [src_rank][local_expert]; ONE stable
         # argsort by local-expert id groups them contiguously (the per-expert
