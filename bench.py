@@ -38,7 +38,7 @@ def parse_args():
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--model", type=str, default="gpt2_1.3b",
                     choices=["gpt2_1.3b", "gpt2_small", "tiny", "llama_8b",
-                             "moe_8x", "moe_tiny"])
+                             "moe_8x", "moe_tiny", "llama_tiny"])
     ap.add_argument("--graph", action="store_true",
                     help="capture the whole train step in one hipGraph "
                          "(world_size==1 only)")
@@ -100,6 +100,9 @@ def main():
     elif args.model == "moe_8x":
         from torchdistpackage_amd.models.moe_model import mixtral_style_8x
         cfg = mixtral_style_8x()
+    elif args.model == "llama_tiny":
+        from torchdistpackage_amd.models.llama import llama_tiny
+        cfg = llama_tiny()
     elif args.model == "moe_tiny":
         from torchdistpackage_amd.models.moe_model import MoEConfig
         cfg = MoEConfig(vocab_size=512, n_layer=2, n_head=2, dim=128,
@@ -163,7 +166,7 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
 
     world = dist.get_world_size() if dist.is_initialized() else 1
     torch.manual_seed(1234)  # same init across ranks (then broadcast anyway)
-    if args.model == "llama_8b":
+    if args.model.startswith("llama"):
         from torchdistpackage_amd.models.llama import LlamaModel
         model = LlamaModel(cfg, device=dev, dtype=dtype)
     elif args.model.startswith("moe"):
