@@ -375,3 +375,20 @@ def test_graphed_step():
     assert torch.isfinite(loss_box["loss"])
     assert not torch.equal(p0, model[0].weight), "params must update"
 
+
+
+def test_tr16_transpose_read_semantics():
+    """ds_read_b64_tr_b16: lane l's element j comes from the 8B-floored
+    address of lane (l&3)+4j of its 16-lane group, plus (l&3) elements
+    (four transposed 4x4 bf16 tiles per group; probe-verified model)."""
+    import torchdistpackage_amd.ops as ops
+    e = ops.ext("probe")
+    for lane_addr in (False, True):
+        r = e.tr16_probe(lane_addr).cpu().numpy()
+        for l in range(64):
+            g0 = (l // 16) * 16
+            for j in range(4):
+                src = g0 + (l & 3) + 4 * j
+                addr_elems = ((src & 15) * 2 if lane_addr else 0) // 8 * 4
+                expect = addr_elems + (l & 3)
+                assert r[l][j] == expect, (l, j, r[l].tolist(), expect)

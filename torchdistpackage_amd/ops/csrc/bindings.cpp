@@ -37,6 +37,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
+torch::Tensor tr16_probe(bool use_lane_addr);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);
@@ -57,6 +58,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe_16x16x32", &mfma_probe_16x16x32);
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
+  m.def("tr16_probe", &tr16_probe);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
 }

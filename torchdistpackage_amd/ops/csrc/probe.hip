@@ -94,3 +94,51 @@ torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b) {
   HIP_CHECK_LAST();
   return d;
 }
+
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 layout probe.  Hardware-verified semantics (this
+// probe, marker fill + two address patterns): within each 16-lane group,
+// lane l receives element j = lds[floor8B(addr(lane (l&3)+4j))/2 + (l&3)]
+// — i.e. the four lanes {c, c+4, c+8, c+12} of a group supply the four
+// 8-byte-aligned ROW addresses of a 4x4 bf16 tile, and every lane of that
+// quad-column reads its (l&3) column across those rows (a free 4x4
+// transpose; four independent tiles per 16-lane group, addresses below
+// 8-byte alignment are floored).  tests/test_ops_gpu.py asserts this.
+// ---------------------------------------------------------------------------
+
+__global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
+                                  int use_lane_addr) {
+  __shared__ __attribute__((aligned(16))) unsigned short lds[256];
+  int tid = threadIdx.x;
+  for (int i = tid; i < 256; i += blockDim.x)
+    lds[i] = (unsigned short)i;          // marker = element index
+  __syncthreads();
+  if (tid < 64) {
+    // per-lane byte address: the ISA form takes a VGPR address; probe both
+    // a uniform base and a per-lane 8B-aligned base to see which operand
+    // model holds (guide: per-lane addr, no internal lane offset beyond
+    // the group gather)
+    // the DS address operand is an LDS-segment byte offset: convert the
+    // generic (flat) shared pointer through address_space(3)
+    unsigned base = (unsigned)(unsigned long long)
+        (__attribute__((address_space(3))) unsigned short*)lds;
+    unsigned addr = base + (use_lane_addr ? (unsigned)((tid & 15) * 2) : 0u);
+    unsigned long long got;
+    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(got) : "v"(addr));
+    unsigned short* g = (unsigned short*)&got;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[tid * 4 + j] = g[j];
+  }
+}
+
+torch::Tensor tr16_probe(bool use_lane_addr) {
+  auto out = torch::zeros({64, 4},
+                          torch::dtype(torch::kInt16).device(torch::kCUDA));
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (unsigned short*)out.data_ptr(),
+                     use_lane_addr ? 1 : 0);
+  HIP_CHECK_LAST();
+  return out;
+}
