@@ -378,17 +378,26 @@ def test_graphed_step():
 
 
 def test_tr16_transpose_read_semantics():
-    """ds_read_b64_tr_b16: lane l's element j comes from the 8B-floored
-    address of lane (l&3)+4j of its 16-lane group, plus (l&3) elements
-    (four transposed 4x4 bf16 tiles per group; probe-verified model)."""
+    """ds_read_b64_tr_b16 hardware model: lane l's element j =
+    lds[floor8B(addr(lane g0+((l>>2)&3)+4j))/2 + (l&3)] — sub-quad q reads
+    the transposed 4x4 tile addressed by lanes {q, q+4, q+8, q+12}."""
     import torchdistpackage_amd.ops as ops
     e = ops.ext("probe")
-    for lane_addr in (False, True):
-        r = e.tr16_probe(lane_addr).cpu().numpy()
+
+    def addr_bytes(lane, mode):
+        if mode == 0:
+            return 0
+        if mode == 1:
+            return (lane & 15) * 2
+        if mode == 2:
+            return (lane & 15) * 8
+        return (lane & 63) * 8
+
+    for mode in (0, 1, 2, 3):
+        r = e.tr16_probe(mode).cpu().numpy()
         for l in range(64):
             g0 = (l // 16) * 16
             for j in range(4):
-                src = g0 + (l & 3) + 4 * j
-                addr_elems = ((src & 15) * 2 if lane_addr else 0) // 8 * 4
-                expect = addr_elems + (l & 3)
-                assert r[l][j] == expect, (l, j, r[l].tolist(), expect)
+                src = g0 + ((l >> 2) & 3) + 4 * j
+                expect = addr_bytes(src, mode) // 8 * 4 + (l & 3)
+                assert r[l][j] == expect, (mode, l, j, r[l].tolist(), expect)

@@ -37,7 +37,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     bool causal, double scale);
 torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
-torch::Tensor tr16_probe(bool use_lane_addr);
+torch::Tensor tr16_probe(long addr_mode);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);

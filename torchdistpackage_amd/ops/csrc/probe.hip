@@ -97,17 +97,21 @@ torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b) {
 
 // ---------------------------------------------------------------------------
 // ds_read_b64_tr_b16 layout probe.  Hardware-verified semantics (this
-// probe, marker fill + two address patterns): within each 16-lane group,
-// lane l receives element j = lds[floor8B(addr(lane (l&3)+4j))/2 + (l&3)]
-// — i.e. the four lanes {c, c+4, c+8, c+12} of a group supply the four
-// 8-byte-aligned ROW addresses of a 4x4 bf16 tile, and every lane of that
-// quad-column reads its (l&3) column across those rows (a free 4x4
-// transpose; four independent tiles per 16-lane group, addresses below
-// 8-byte alignment are floored).  tests/test_ops_gpu.py asserts this.
+// probe, marker fill + four address patterns — uniform, quad-aliased,
+// per-lane distinct, cross-group distinct):
+//
+//   lane l, element j = lds[ floor8B(addr(lane g0 + ((l>>2)&3) + 4j))/2
+//                            + (l&3) ]        (g0 = 16*(l/16))
+//
+// i.e. within each 16-lane group, sub-quad q=(l>>2)&3 reads the 4x4 bf16
+// tile whose four ROW addresses come from lanes {q, q+4, q+8, q+12},
+// transposed (lane column = l&3); four independent tiles per group, 16
+// distinct columns; address bits below 8-byte alignment are floored.
+// tests/test_ops_gpu.py asserts this model against all four patterns.
 // ---------------------------------------------------------------------------
 
 __global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
-                                  int use_lane_addr) {
+                                  int addr_mode) {
   __shared__ __attribute__((aligned(16))) unsigned short lds[256];
   int tid = threadIdx.x;
   for (int i = tid; i < 256; i += blockDim.x)
@@ -122,7 +126,11 @@ __global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
     // generic (flat) shared pointer through address_space(3)
     unsigned base = (unsigned)(unsigned long long)
         (__attribute__((address_space(3))) unsigned short*)lds;
-    unsigned addr = base + (use_lane_addr ? (unsigned)((tid & 15) * 2) : 0u);
+    unsigned off = 0;
+    if (addr_mode == 1) off = (unsigned)((tid & 15) * 2);
+    else if (addr_mode == 2) off = (unsigned)((tid & 15) * 8);
+    else if (addr_mode == 3) off = (unsigned)((tid & 63) * 8);
+    unsigned addr = base + off;
     unsigned long long got;
     asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                  : "=v"(got) : "v"(addr));
@@ -132,13 +140,12 @@ __global__ void tr16_probe_kernel(unsigned short* __restrict__ out,
   }
 }
 
-torch::Tensor tr16_probe(bool use_lane_addr) {
+torch::Tensor tr16_probe(long addr_mode) {
   auto out = torch::zeros({64, 4},
                           torch::dtype(torch::kInt16).device(torch::kCUDA));
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
-                     (unsigned short*)out.data_ptr(),
-                     use_lane_addr ? 1 : 0);
+                     (unsigned short*)out.data_ptr(), (int)addr_mode);
   HIP_CHECK_LAST();
   return out;
 }
