@@ -965,6 +965,149 @@ __global__ void attn_bwd_dq_lite_kernel(
 
 }  // namespace
 
+// ===========================================================================
+// dq-lite-tr (default; TDPA_NO_TR16 falls back): same math but the K^T
+// A-fragments come from ds_read_b64_tr_b16 over a BLOCKED natural image
+// kimg[d/16][key][16] instead of a transposed tile built by a 64-write
+// scalar scatter.  Address derivation from the probe-verified gather model
+// (see probe.hip): lane l elem j = lds[floor8B(addr(lane ((l>>2)&3)+4j))/2
+// + (l&3)]; required element for the A-frag is
+//   (d>>4)*1024 + key*16 + (d&15),   d = ds*32 + (l&31)
+// which splits into a per-lane base
+//   base(k) = 2*[ ((k>>4)&1)*1024 + ((k>>2)&3)*16 + 4*(k&3) + (k>>5)*128 ]
+// plus a uniform immediate offset ds*4096 + (kt*32+kc*16+sub*4)*32 bytes.
+// ===========================================================================
+
+namespace {
+
+template <bool CAUSAL>
+__launch_bounds__(512)
+__global__ void attn_bwd_dq_lite_tr_kernel(
+    const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ dsb,
+    unsigned short* __restrict__ dq,
+    Strides2 ks, Strides2 dqs,
+    int B, int H, int S, long spad, int q_per_kv) {
+  __shared__ __attribute__((aligned(16))) unsigned short kimg[2][8 * KV * 16];
+
+  const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
+  const int hkv = hh / q_per_kv;
+  const int qtile = CAUSAL ? (gridDim.x - 1 - blockIdx.x) : blockIdx.x;
+  const int qbase = qtile * QT2;
+  if (qbase >= S) return;
+  const int tid = threadIdx.x;
+  const int wid = tid / WAVE;
+  const int lane = tid % WAVE;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;
+
+  const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
+  const int qrow = qbase + wid * QW + l31;
+  const unsigned short* dsp = dsb +
+      ((long)bh * S + (qrow < S ? qrow : S - 1)) * spad;
+
+  // per-lane tr16 base (bytes, within kimg[0]); buffer 1 adds 16 KiB
+  const unsigned lds_base = (unsigned)(unsigned long long)
+      (__attribute__((address_space(3))) unsigned short*)&kimg[0][0];
+  const unsigned tr_base = lds_base + 2u * (((lane >> 4) & 1) * 1024u +
+                                            ((lane >> 2) & 3) * 16u +
+                                            (lane & 3) * 4u +
+                                            (lane >> 5) * 128u);
+
+  f32x16 acc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) acc[ds] = (f32x16)(0.f);
+
+  const int kv_end = CAUSAL ? min(S, qbase + QT2) : S;
+
+  bf16x8_v k_reg[2];
+  auto load_regs = [&](int kt0) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int gkey = kt0 + idx / D2;
+      k_reg[c] = gkey < S ? pack8v(kp + (long)gkey * ks.s + idx % D2)
+                          : (bf16x8_v)(__bf16)0.f;
+    }
+  };
+  auto stage_img = [&](int b) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = tid * 8 + c * 4096;
+      int key = idx / D2;
+      int col = idx % D2;
+      // blocked natural image: [col/16][key][col%16], b128 writes
+      *(bf16x8_v*)&kimg[b][(col >> 4) * (KV * 16) + key * 16 + (col & 15)] =
+          k_reg[c];
+    }
+  };
+  load_regs(0);
+  stage_img(0);
+  if (KV < kv_end) load_regs(KV);
+  __syncthreads();
+
+  int buf = 0;
+  for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
+    const unsigned abase = tr_base + (unsigned)(buf * 2 * 8 * KV * 16);
+    bf16x8_v db[2][2];
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+        db[kt][kc] = pack8v(dsp + kt0 + kt * 32 + kc * 16 + hi * 8);
+
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) {
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          // K^T A-frag via two 4-key transpose reads (keys sub*4)
+          unsigned long long lo, hic;
+          const unsigned off = (unsigned)(ds * 4096 +
+                                          (kt * 32 + kc * 16) * 32);
+          asm volatile(
+              "ds_read_b64_tr_b16 %0, %2 offset:%3\n\t"
+              "ds_read_b64_tr_b16 %1, %2 offset:%4\n\t"
+              "s_waitcnt lgkmcnt(0)"
+              : "=v"(lo), "=v"(hic)
+              : "v"(abase + off), "i"(0), "i"(128));
+          bf16x8_v a_kt;
+          ((unsigned long long*)&a_kt)[0] = lo;
+          ((unsigned long long*)&a_kt)[1] = hic;
+          acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_kt, db[kt][kc], acc[ds], 0, 0, 0);
+        }
+      }
+    }
+
+    if (kt0 + KV < kv_end) {
+      stage_img(buf ^ 1);
+      if (kt0 + 2 * KV < kv_end) load_regs(kt0 + 2 * KV);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  if (qrow < S) {
+    unsigned short* dqp = dq + bb * dqs.b + hh * dqs.h + (long)qrow * dqs.s;
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+      for (int blk = 0; blk < 4; ++blk) {
+        int d0 = ds * 32 + 8 * blk + 4 * hi;
+        unsigned short out4[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          out4[j] = f2bf(acc[ds][4 * blk + j]);
+        *(ushort4*)(dqp + d0) = *(ushort4*)out4;
+      }
+    }
+  }
+}
+
+}  // namespace
 // Combined backward: dv pass, dk pass storing dS, dq-lite consuming it.
 // TDPA_DQ_RECOMPUTE falls back to the independent recomputing dq kernel
 // (and no dS workspace) for A/B and as an escape hatch.
@@ -993,11 +1136,20 @@ void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   if (causal) L_KV2(true, true, dsp, spad);
   else L_KV2(false, true, dsp, spad);
 #undef L_KV2
+  static const bool no_tr16 = std::getenv("TDPA_NO_TR16") != nullptr;
 #define L_DQL(CC)                                                             \
-  hipLaunchKernelGGL((attn_bwd_dq_lite_kernel<CC>), grid, block, 0, stream,   \
-                     (const unsigned short*)k.data_ptr(), dsp,                \
-                     (unsigned short*)dq.data_ptr(), get(k), get(dq),         \
-                     B, H, S, spad, q_per_kv)
+  do {                                                                        \
+    if (no_tr16)                                                              \
+      hipLaunchKernelGGL((attn_bwd_dq_lite_kernel<CC>), grid, block, 0,       \
+                         stream, (const unsigned short*)k.data_ptr(), dsp,    \
+                         (unsigned short*)dq.data_ptr(), get(k), get(dq),     \
+                         B, H, S, spad, q_per_kv);                            \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_bwd_dq_lite_tr_kernel<CC>), grid, block, 0,    \
+                         stream, (const unsigned short*)k.data_ptr(), dsp,    \
+                         (unsigned short*)dq.data_ptr(), get(k), get(dq),     \
+                         B, H, S, spad, q_per_kv);                            \
+  } while (0)
   if (causal) L_DQL(true); else L_DQL(false);
 #undef L_DQL
   HIP_CHECK_LAST();
