@@ -10,6 +10,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires an AMD GPU (run on MI355X via gpurun)")
+    config.addinivalue_line(
+        "markers", "slow: long-running contract test (8-rank subprocess)")
 
 
 def pytest_collection_modifyitems(config, items):
