@@ -144,9 +144,11 @@ def test_flash_attention_fwd(shape, causal):
 
 
 @pytest.mark.parametrize("causal", [True, False])
-def test_flash_attention_bwd(causal):
+@pytest.mark.parametrize("shape", [
+    (2, 2, 256, 128), (1, 2, 100, 128), (1, 1, 300, 128), (2, 2, 192, 64)])
+def test_flash_attention_bwd(shape, causal):
     torch.manual_seed(5)
-    B, H, S, D = 2, 2, 256, 128
+    B, H, S, D = shape
     q = torch.randn(B, H, S, D, dtype=torch.bfloat16, device=_dev(),
                     requires_grad=True)
     k = torch.randn_like(q, requires_grad=True)
