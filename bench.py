@@ -216,7 +216,7 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         if isinstance(model, NaiveDdp):
             model.reduce_gradients()
             from torchdistpackage_amd.ddp import moe_dp_iter_step
-            if args.model == "moe_8x":
+            if args.model.startswith("moe"):
                 moe_dp_iter_step()
         if tp > 1:
             allreduce_sequence_parallel_grads(
