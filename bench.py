@@ -180,6 +180,10 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         model = GPT2Model(cfg, device=dev, dtype=dtype)
 
     ema = None
+    if args.zero and args.model.startswith("moe"):
+        raise SystemExit("--zero shards optimizer state over the dp group, "
+                         "which is wrong for expert-parallel params (each EP "
+                         "rank owns different experts); use NaiveDdp+MoeDP")
     if args.zero:
         # ZeRO owns grad reduction — model stays unwrapped (wrapping with
         # NaiveDdp too would leave BOTH hook sets firing)
