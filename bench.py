@@ -44,7 +44,8 @@ def parse_args():
                          "(world_size==1 only)")
     ap.add_argument("--zero", action="store_true",
                     help="use Bf16ZeroOptimizer (hybrid node-local shard) + "
-                         "sharded EMA instead of plain FusedAdamW")
+                         "sharded EMA instead of plain FusedAdamW "
+                         "(dp/tp path only; ignored when pp > 1)")
     ap.add_argument("--micro-batches", type=int, default=8,
                     help="micro-batches per step when PP is active")
     return ap.parse_args()
@@ -186,15 +187,25 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
                          "rank owns different experts); use NaiveDdp+MoeDP")
     if args.zero:
         # ZeRO owns grad reduction — model stays unwrapped (wrapping with
-        # NaiveDdp too would leave BOTH hook sets firing)
+        # NaiveDdp too would leave BOTH hook sets firing).
+        # Sharding must stay WITHIN a set of ranks holding IDENTICAL params:
+        # the intra-node hybrid group only when the node is pure-dp; with
+        # tp/pp active only the dp group is replicated (an all-gather over
+        # mixed tp/pp ranks would overwrite each shard's weights).
         from torchdistpackage_amd import Bf16ZeroOptimizer, ShardedEMA, \
             setup_node_groups
-        node_group = setup_node_groups(num_per_node=8) if world > 1 else None
+        if world > 1 and tp > 1:   # this path runs with pp == 1
+            shard_group = tpc.get_group("data")
+        elif world > 1:
+            shard_group = setup_node_groups(num_per_node=min(8, world))
+        else:
+            shard_group = None
         inner_opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
-        opt = Bf16ZeroOptimizer(inner_opt, group=node_group,
+        opt = Bf16ZeroOptimizer(inner_opt, group=shard_group,
                                 grad_group=tpc.get_group("data")
                                 if world > 1 else None, stage2=True)
-        ema = ShardedEMA(model, decay=0.999)
+        ema = ShardedEMA(model, decay=0.999,
+                         group=tpc.get_group("data") if world > 1 else None)
     else:
         if world > 1 and dp > 1:
             model = NaiveDdp(model, group=tpc.get_group("data"))
