@@ -20,7 +20,8 @@ from torchdistpackage_amd.ops.optim import FusedAdamW
 def main():
     info = setup_distributed()
     tpc.setup_process_groups([("data", info["world_size"])])
-    node_group = setup_node_groups(num_per_node=8)
+    node_group = setup_node_groups(
+        num_per_node=min(8, info["world_size"]))
     fix_rand(info["rank"])
     dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
     dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
