@@ -22,11 +22,14 @@ def main():
     tpc.setup_process_groups([("data", info["world_size"])])
     node_group = setup_node_groups(
         num_per_node=min(8, info["world_size"]))
-    fix_rand(info["rank"])
+    # seed IDENTICALLY for model init so DP replicas start from the same
+    # weights (ZeRO never broadcasts params); re-seed per-rank for data below
+    fix_rand(0)
     dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
     dtype = torch.bfloat16 if dev.type == "cuda" else torch.float32
 
     model = LlamaModel(llama_tiny(), device=dev, dtype=dtype)
+    fix_rand(info["rank"])
     inner = FusedAdamW(model.parameters(), lr=1e-4)
     # hybrid: shard optimizer state inside the node, reduce grads over all DP
     opt = Bf16ZeroOptimizer(inner, group=node_group,

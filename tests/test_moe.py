@@ -242,3 +242,24 @@ def _bench_order_regression(rank, world_size):
 
 def test_bench_order_regression():
     run_distributed(_bench_order_regression, world_size=2)
+
+
+def _moe_ep1_in_multirank_world(rank, world_size):
+    """ADVICE r01 regression: ExpertParallelMoE with ep_group=None (no moe
+    groups built) inside a multi-rank job must NOT issue the combine
+    all-to-all over WORLD (it crashed / mis-split before the guard)."""
+    from torchdistpackage_amd.moe.layer import ExpertParallelMoE
+
+    torch.manual_seed(5)
+    moe = ExpertParallelMoE(dim=16, num_experts=4, top_k=2, ep_group=None)
+    assert moe.ep_size == 1
+    x = torch.randn(3, 7, 16)
+    y = moe(x)
+    assert y.shape == x.shape
+    assert torch.isfinite(y).all()
+    y.sum().backward()
+    return True
+
+
+def test_moe_ep1_guard_world2():
+    run_distributed(_moe_ep1_in_multirank_world, world_size=2)

@@ -127,3 +127,49 @@ def test_zero_with_fused_adamw():
 
 def test_zero_with_fused_adamw_world2():
     run_distributed(_zero_with_fused_adamw, world_size=2)
+
+
+def _zero_tp_sp_grads(rank, world_size):
+    """ADVICE r01 regression: ZeRO + TP/SP composition.  SP-tagged params'
+    grads are fixed up by an all-reduce over TP AFTER backward; ZeRO must not
+    snapshot them into buckets before that (the bucketed copy silently
+    dropped the TP all-reduce)."""
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+    from torchdistpackage_amd.parallel.tensor.tp_utils import (
+        set_tp_group, allreduce_sequence_parallel_grads)
+
+    tpc.setup_process_groups([("data", 2), ("tensor", 2)])
+    set_tp_group(tpc.get_group("tensor"))
+    dp_rank = tpc.get_dp_rank()
+    tp_rank = tpc.get_tp_rank()
+
+    torch.manual_seed(7)
+    model = nn.Linear(8, 8, bias=True)
+    model.bias.sequence_parallel_param = True  # plays the SP LayerNorm role
+
+    inner = torch.optim.Adam(model.parameters(), lr=1e-3)
+    zopt = Bf16ZeroOptimizer(inner, group=tpc.get_group("data"),
+                             grad_group=tpc.get_group("data"))
+
+    # deterministic per-rank grads: d(loss)/dw = a per element, d/db = c
+    a = 1.0 + dp_rank
+    c = 1.0 + 2 * dp_rank + 10 * tp_rank
+    (model.weight.sum() * a + model.bias.sum() * c).backward()
+    allreduce_sequence_parallel_grads(model)
+    zopt._finish_reduction()
+
+    # expected: weight grad = avg_dp(a) = 1.5;
+    # bias grad = avg_dp(sum_tp(c)) = ((12+0)+(12+4))/2 = 14
+    for i in zopt._my_idx:
+        mp = zopt._master_params[i]
+        p = zopt._params[i]
+        expect = 1.5 if p is model.weight else 14.0
+        assert mp.grad is not None
+        assert torch.allclose(mp.grad, torch.full_like(mp, expect)), \
+            (rank, p.shape, mp.grad.flatten()[:4], expect)
+    return True
+
+
+def test_zero_tp_sp_grads_world4():
+    run_distributed(_zero_tp_sp_grads, world_size=4)

@@ -119,6 +119,7 @@ class MoEDP:
             self._pending.clear()
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data, op=dist.ReduceOp.AVG,
                                     group=self.group)
                     b.reset()
@@ -129,14 +130,16 @@ class MoEDP:
             self._works.clear()
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data, op=dist.ReduceOp.SUM,
                                     group=self.group)
                     b.data.div_(self._world())
                     b.reset()
         for b in self._buckets:
-            for p, v in zip(b.params, b.views):
-                if p.grad is not None:
+            for p, v, pu in zip(b.params, b.views, b.pushed):
+                if pu and p.grad is not None:
                     p.grad.copy_(v)
+            b.clear_pushed()
 
 
 # module-level convenience API, reference-compatible

@@ -60,16 +60,32 @@ class GradBucket:
             for o, p in zip(self.offsets, params)
         ]
         self.ready = 0
+        # which members arrived THIS comm iteration (cleared by the wrapper
+        # after writeback, not by reset()): a partial-bucket flush must not
+        # reduce stale data left in un-pushed views from a prior iteration
+        self.pushed = [False] * len(params)
         self.reduced_event: Optional[torch.cuda.Event] = None
 
     def push(self, idx: int, grad: torch.Tensor) -> bool:
         """Copy grad into its view; True when every member has arrived."""
         self.views[idx].copy_(grad)
+        self.pushed[idx] = True
         self.ready += 1
         return self.ready == len(self.params)
 
     def reset(self):
         self.ready = 0
+
+    def clear_pushed(self):
+        for i in range(len(self.pushed)):
+            self.pushed[i] = False
+
+    def zero_unpushed(self):
+        """Zero views whose param produced no grad this iteration so a
+        partial flush contributes zeros instead of stale values."""
+        for i, v in enumerate(self.views):
+            if not self.pushed[i]:
+                v.zero_()
 
 
 class NaiveDdp(nn.Module):
@@ -257,17 +273,21 @@ class NaiveDdp(nn.Module):
                 cur.wait_event(ev)
             self._pending_events.clear()
             # flush any bucket that never filled (params with no grad this
-            # iter would leave it partial — reduce what's there)
+            # iter leave it partial): un-pushed views are zeroed first so the
+            # collective contributes zeros, never stale values from an
+            # earlier iteration
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data,
                                     op=dist.ReduceOp.AVG if self._avg()
                                     else dist.ReduceOp.SUM, group=self.group)
                     b.reset()
             for b in self._buckets:
-                for p, v in zip(b.params, b.views):
-                    if p.grad is not None:
+                for p, v, pu in zip(b.params, b.views, b.pushed):
+                    if pu and p.grad is not None:
                         p.grad.copy_(v)
+                b.clear_pushed()
         else:
             for work, b in self._works:
                 work.wait()
@@ -276,15 +296,17 @@ class NaiveDdp(nn.Module):
             self._works.clear()
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data, op=dist.ReduceOp.SUM,
                                     group=self.group)
                     if self._avg():
                         b.data.div_(self._world())
                     b.reset()
             for b in self._buckets:
-                for p, v in zip(b.params, b.views):
-                    if p.grad is not None:
+                for p, v, pu in zip(b.params, b.views, b.pushed):
+                    if pu and p.grad is not None:
                         p.grad.copy_(v)
+                b.clear_pushed()
 
     def remove_hooks(self):
         """Detach all grad hooks (e.g. to re-wrap the module elsewhere)."""

@@ -22,6 +22,13 @@ MI355X-first design decisions:
 - 288 GB HBM3E sizing: default bucket is 100 MiB (larger messages amortize
   RCCL launch overhead; the xGMI per-link bound makes few-large better than
   many-small), and master/optimizer state lives wholly on-device.
+- TP/SP composition: params tagged ``sequence_parallel_param`` (LayerNorm /
+  RMSNorm inside SP regions) are NOT bucketed during backward.  Their grads
+  need an all-reduce over the TP group (``allreduce_sequence_parallel_grads``,
+  called by the user after backward, before ``step()``) and bucketing would
+  snapshot the grad BEFORE that fix-up; instead they are reduced from
+  ``p.grad`` in one small flat collective at ``step()`` time, after the SP
+  all-reduce has run.
 """
 
 from __future__ import annotations
@@ -115,9 +122,11 @@ class Bf16ZeroOptimizer:
         self._works = []
         self._buckets: List[GradBucket] = []
         self._param_bucket: Dict[int, tuple] = {}
+        self._late_idx: List[int] = []
         self._build_buckets()
         self._hooks = [p.register_post_accumulate_grad_hook(self._on_grad_ready)
-                       for p in self._params]
+                       for p in self._params
+                       if id(p) in self._param_bucket]
 
         # gather buffers for the post-step param all-gather
         self._max_shard = max(
@@ -129,6 +138,11 @@ class Bf16ZeroOptimizer:
     def _build_buckets(self):
         cur, cur_bytes = [], 0
         for p in reversed(self._params):
+            if getattr(p, "sequence_parallel_param", False):
+                # reduced late, from p.grad, after the user's SP all-reduce
+                # over the TP group has fixed the grad up (see module doc)
+                self._late_idx.append(self._idx_of[id(p)])
+                continue
             nbytes = _align(p.numel()) * p.element_size()
             if cur and cur_bytes + nbytes > self.bucket_cap:
                 self._close(cur)
@@ -197,6 +211,7 @@ class Bf16ZeroOptimizer:
             self._pending_events.clear()
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data, op=dist.ReduceOp.AVG,
                                     group=self.grad_group)
                     b.reset()
@@ -207,13 +222,47 @@ class Bf16ZeroOptimizer:
             self._works.clear()
             for b in self._buckets:
                 if b.ready > 0:
+                    b.zero_unpushed()
                     dist.all_reduce(b.data, op=dist.ReduceOp.SUM,
                                     group=self.grad_group)
                     b.data.div_(self._grad_world())
                     b.reset()
-        # copy (cast) owned grads into master grads; optionally free the rest
+        # late (SP-tagged) params: reduce from p.grad NOW — after the user's
+        # allreduce_sequence_parallel_grads over TP has fixed the grads up
+        # (AVG over dp and SUM over tp commute, so the order is free)
+        late = [self._params[i] for i in self._late_idx
+                if self._params[i].grad is not None]
+        if late:
+            flat = torch.cat([p.grad.reshape(-1) for p in late])
+            if self._use_gpu:
+                dist.all_reduce(flat, op=dist.ReduceOp.AVG,
+                                group=self.grad_group)
+            else:
+                dist.all_reduce(flat, op=dist.ReduceOp.SUM,
+                                group=self.grad_group)
+                flat.div_(self._grad_world())
+            off = 0
+            for p in late:
+                p.grad.copy_(flat[off:off + p.numel()].view_as(p))
+                off += p.numel()
+        for i in self._late_idx:
+            p = self._params[i]
+            if p.grad is None:
+                continue
+            if self._owner[i] == self.rank:
+                mp = self._master_params[i]
+                if mp.grad is None:
+                    mp.grad = torch.empty_like(mp)
+                mp.grad.copy_(p.grad)
+            if self.stage2:
+                p.grad = None
+        # copy (cast) owned grads into master grads; optionally free the rest.
+        # Only params whose hook pushed THIS iteration: a never-fired param's
+        # view holds a previous iteration's reduced grad (stale).
         for b in self._buckets:
-            for p, v in zip(b.params, b.views):
+            for p, v, pu in zip(b.params, b.views, b.pushed):
+                if not pu:
+                    continue
                 i = self._idx_of[id(p)]
                 if self._owner[i] == self.rank:
                     mp = self._master_params[i]
@@ -222,6 +271,7 @@ class Bf16ZeroOptimizer:
                     mp.grad.copy_(v.view(p.shape))
                 if self.stage2:
                     p.grad = None
+            b.clear_pushed()
 
     # ------------------------------------------------------------------
 
@@ -236,11 +286,12 @@ class Bf16ZeroOptimizer:
 
     @torch.no_grad()
     def _clip_master_grads(self, max_norm: float):
+        from ..ops import l2norm_sq
         grads = [self._master_params[i].grad for i in self._my_idx
                  if self._master_params[i].grad is not None]
         if grads:
             local_sq = torch.stack(
-                [g.float().pow(2).sum() for g in grads]).sum()
+                [l2norm_sq(g.reshape(-1)) for g in grads]).sum()
         else:
             local_sq = torch.zeros((), device=self._master_flat.device)
         if dist.is_initialized() and self.world > 1:
@@ -264,13 +315,19 @@ class Bf16ZeroOptimizer:
             return
         dtype = self._params[0].dtype
         dev = self._params[0].device
-        send = torch.zeros(self._max_shard, dtype=dtype, device=dev)
+        # persistent send/recv buffers: a fresh (max_shard * world) alloc per
+        # step churns the allocator for nothing
+        if getattr(self, "_gather_send", None) is None:
+            self._gather_send = torch.zeros(self._max_shard, dtype=dtype,
+                                            device=dev)
+            self._gather_recv = torch.empty(self._max_shard * self.world,
+                                            dtype=dtype, device=dev)
+        send, recv = self._gather_send, self._gather_recv
         off = 0
         for i in self._my_idx:
             n = self._params[i].numel()
             send[off:off + n].copy_(self._params[i].reshape(-1))
             off += n
-        recv = torch.empty(self._max_shard * self.world, dtype=dtype, device=dev)
         dist.all_gather_into_tensor(recv, send, group=self.group)
         for r, part in enumerate(self._parts):
             if r == self.rank:

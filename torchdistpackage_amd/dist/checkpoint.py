@@ -102,9 +102,18 @@ def save_checkpoint(directory: str, step: int, model: torch.nn.Module,
                                  _is_save_rank()):
             torch.save(full, os.path.join(
                 directory, f"ema_step{step}{get_mp_ckpt_suffix()}.pth"))
+    # all shard files must exist before 'latest' moves: barrier first, then
+    # rank 0 publishes via fsync'd tmp file + atomic rename — a crash mid-save
+    # can never leave 'latest' pointing at an incomplete checkpoint set
+    if dist.is_initialized():
+        dist.barrier()
     if not dist.is_initialized() or dist.get_rank() == 0:
-        with open(os.path.join(directory, "latest"), "w") as f:
+        tmp = os.path.join(directory, "latest.tmp")
+        with open(tmp, "w") as f:
             f.write(str(step))
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, os.path.join(directory, "latest"))
     if dist.is_initialized():
         dist.barrier()
 

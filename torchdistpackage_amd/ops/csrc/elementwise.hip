@@ -340,6 +340,10 @@ __global__ void multi_adamw_kernel(const int* __restrict__ cpid,
   float* master = (float*)mptrs[pid];  // per-param master (MAY alias param)
   void* pp = (void*)pptrs[pid];
   const void* gp = (const void*)gptrs[pid];
+  // grad-None params are SKIPPED entirely (no decay, no moment update) —
+  // stock torch.optim semantics; substituting grad=0 silently decayed
+  // frozen params toward zero.
+  if (gp == nullptr) return;
   // HBM-bound: ~26 B/element of read+write traffic, so dwordx4 accesses are
   // what decides the rate.  All pointers/offsets are 16B-aligned in practice
   // (torch allocations + 64Ki chunking); fall back to scalars otherwise.

@@ -42,6 +42,7 @@ class _FlatGroup:
         for p in params:
             offs.append(off)
             off += p.numel()
+        self.offs = offs
         self.exp_avg = torch.zeros(n, dtype=torch.float32, device=dev)
         self.exp_avg_sq = torch.zeros(n, dtype=torch.float32, device=dev)
         self.master_views = []
@@ -145,9 +146,22 @@ class FusedAdamW(torch.optim.Optimizer):
             grads = [p.grad if p.grad is not None
                      else torch.zeros_like(p) for p in fg.params]
             torch._foreach_copy_(fg.grad_views, grads)
+            # grad-None params must be SKIPPED (stock torch semantics): save
+            # their flat segments and restore after the whole-flat update
+            saves = []
+            for i, p in enumerate(fg.params):
+                if p.grad is None:
+                    o, n = fg.offs[i], p.numel()
+                    saves.append((o, n, fg.master[o:o + n].clone(),
+                                  fg.exp_avg[o:o + n].clone(),
+                                  fg.exp_avg_sq[o:o + n].clone()))
             fused_adamw_(fg.master, fg.grad32, fg.exp_avg, fg.exp_avg_sq,
                          self._step, lr, beta1, beta2, group["eps"],
                          group["weight_decay"])
+            for o, n, ms, es, vs in saves:
+                fg.master[o:o + n].copy_(ms)
+                fg.exp_avg[o:o + n].copy_(es)
+                fg.exp_avg_sq[o:o + n].copy_(vs)
             torch._foreach_copy_(fg.params, fg.master_views)
 
     def zero_grad(self, set_to_none: bool = True):
