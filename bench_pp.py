@@ -26,13 +26,8 @@ class PPStage(nn.Module):
         self.is_last = is_last
         self.num_mb = num_microbatches
         self.tp_size = tp_size
-        self.labels_full = None     # set per iteration by the caller
-        self._mb = 0
 
-    def reset_iter(self):
-        self._mb = 0
-
-    def forward(self, inp):
+    def forward(self, inp, labels=None):
         from torchdistpackage_amd.parallel.tensor import (
             set_sequence_parallel_attr, is_sequence_parallel,
             gather_from_sequence_parallel_region)
@@ -49,9 +44,6 @@ class PPStage(nn.Module):
             x = layer(x)
         if self.is_last:
             logits = x  # head output (B_mb, S, V)
-            mb_size = self.labels_full.shape[0] // self.num_mb
-            labels = self.labels_full.narrow(0, self._mb * mb_size, mb_size)
-            self._mb += 1
             from torchdistpackage_amd.ops import cross_entropy_loss
             loss = cross_entropy_loss(logits.transpose(0, 1),
                                       labels.transpose(0, 1)) / self.num_mb
@@ -93,18 +85,18 @@ def run_pp_bench(args, cfg, dev, dtype, dp, pp, tp):
     B_total = args.batch * num_mb
     x = torch.randint(0, cfg.vocab_size, (B_total, args.seq),
                       generator=g).to(dev)
-    stage.labels_full = x
 
-    def fwd_fn(stage_in):
+    def fwd_fn(stage_in, labels=None):
         # first stage receives the sliced token micro-batch, later stages
-        # the previous stage's activation shard
-        return stage_ddp(stage_in) if isinstance(stage_ddp, NaiveDdp) \
-            else stage(stage_in)
+        # the previous stage's activation shard; the last stage gets its
+        # label micro-batch from the scheduler (extra_inputs)
+        m = stage_ddp if isinstance(stage_ddp, NaiveDdp) else stage
+        return m(stage_in, labels=labels)
 
     def step():
-        stage.reset_iter()
         forward_backward(fwd_fn, inputs=x if stage.is_first else None,
-                         num_microbatches=num_mb)
+                         num_microbatches=num_mb,
+                         extra_inputs=x if stage.is_last else None)
         if isinstance(stage_ddp, NaiveDdp):
             stage_ddp.reduce_gradients()
         if tp > 1:

@@ -177,3 +177,59 @@ def _pp_with_ddp(rank, world_size):
 
 def test_pp2_dp2_composition():
     run_distributed(_pp_with_ddp, world_size=4)
+
+
+def _pp_extra_inputs(rank, world_size, num_microbatches=4):
+    """Scheduler-native extra per-stage inputs (reference
+    pipeline_sched.py:6-33): last-stage labels sliced into micro-batches by
+    forward_backward itself — no mutable stage-state side channel."""
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.pipeline import (forward_backward,
+                                                        partition_uniform)
+
+    tpc.setup_process_groups([("pipe", world_size)])
+    depth, dim, B = 6, 32, 8
+    full = _make_layers(seed=42, depth=depth, dim=dim)
+    ref = copy.deepcopy(full)
+
+    parts = partition_uniform(depth, world_size)
+    s, e = parts[rank]
+    stage = nn.Sequential(*list(full)[s:e])
+
+    torch.manual_seed(7)
+    x = torch.randn(B, dim)
+    labels = torch.randn(B, dim)
+    is_last = tpc.is_last_in_pipeline_group()
+
+    seen_mb = []
+
+    def fwd_fn(stage_in, extra=None):
+        out = stage(stage_in)
+        if is_last:
+            assert extra is not None and extra.shape[0] == B // num_microbatches
+            seen_mb.append(extra)
+            return (out - extra).pow(2).mean() / num_microbatches
+        return out
+
+    forward_backward(fwd_fn, inputs=x, num_microbatches=num_microbatches,
+                     extra_inputs=labels if is_last else None)
+
+    if is_last:
+        # micro-batches must arrive in order and tile the full labels
+        assert torch.equal(torch.cat(seen_mb, dim=0), labels)
+
+    (ref(x) - labels).pow(2).mean().backward()
+    ref_stage = nn.Sequential(*list(ref)[s:e])
+    for (n, p), (rn, rp) in zip(stage.named_parameters(),
+                                ref_stage.named_parameters()):
+        assert p.grad is not None
+        assert torch.allclose(p.grad, rp.grad, atol=1e-5), (rank, n)
+    return True
+
+
+def test_1f1b_extra_inputs_pp2():
+    run_distributed(_pp_extra_inputs, world_size=2)
+
+
+def test_1f1b_extra_inputs_pp3():
+    run_distributed(_pp_extra_inputs, world_size=3)

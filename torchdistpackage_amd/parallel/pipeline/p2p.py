@@ -106,12 +106,17 @@ def _communicate(send_prev: Optional[torch.Tensor] = None,
 
     if ops:
         reqs = dist.batch_isend_irecv(ops)
+        # Work.wait() on ProcessGroupNCCL blocks the CURRENT STREAM on the
+        # RCCL stream's completion event (no host/device-wide sync): after
+        # this loop, consuming the recv buffers on the compute stream is
+        # correctly ordered, and NaiveDdp's overlapped side-stream reduce is
+        # left running.  The reference instead hard-syncs the whole device
+        # after every batched p2p (comm.py:322-327) — a documented crutch
+        # that serializes the pipe; its race does not exist under torch 2.x
+        # coalesced batch_isend_irecv.  On gloo, wait() blocks the host
+        # until delivery, which is the CPU-test equivalent.
         for r in reqs:
             r.wait()
-        if dev.type == "cuda":
-            # guard against the batch_isend_irecv completion race the
-            # reference documents (comm.py:322-327)
-            torch.cuda.synchronize()
     return recv_prev, recv_next
 
 

@@ -51,7 +51,8 @@ def forward_backward(fwd_fn: Callable,
                      forward_only: bool = False,
                      grad_scaler=None,
                      dtype: Optional[torch.dtype] = None,
-                     return_losses: bool = False):
+                     return_losses: bool = False,
+                     extra_inputs=None):
     """Run one 1F1B iteration over ``num_microbatches``.
 
     Args:
@@ -64,6 +65,13 @@ def forward_backward(fwd_fn: Callable,
             only; other stages may pass None).
         bwd_fn: optional ``bwd_fn(output, grad_output)`` override.
         forward_only: run pipelined inference instead (no backward).
+        extra_inputs: optional full-batch tensor(s) available to THIS stage
+            (e.g. labels on the last stage), sliced along dim0 into
+            micro-batches by the scheduler and passed as
+            ``fwd_fn(stage_input, extra_mb)`` — the reference slices extra
+            per-stage inputs the same way (pipeline_sched.py:6-33); without
+            this, callers had to side-channel labels through mutable stage
+            state.
     Returns the last micro-batch's output (last stage) — losses list if
     ``return_losses``.
     """
@@ -77,7 +85,12 @@ def forward_backward(fwd_fn: Callable,
     if pp_size == 1:
         outs = []
         for i in range(num_microbatches):
-            out = fwd_fn(_slice_microbatch(inputs, i, num_microbatches))
+            mb_in = _slice_microbatch(inputs, i, num_microbatches)
+            if extra_inputs is not None:
+                out = fwd_fn(mb_in, _slice_microbatch(
+                    extra_inputs, i, num_microbatches))
+            else:
+                out = fwd_fn(mb_in)
             if not forward_only:
                 bwd(out, None)
             outs.append(out.detach() if torch.is_tensor(out) else out)
@@ -100,7 +113,11 @@ def forward_backward(fwd_fn: Callable,
         nonlocal fwd_idx, out_shape, out_dtype
         if is_first:
             stage_in = _slice_microbatch(inputs, fwd_idx, num_microbatches)
-        out = fwd_fn(stage_in)
+        if extra_inputs is not None:
+            out = fwd_fn(stage_in, _slice_microbatch(
+                extra_inputs, fwd_idx, num_microbatches))
+        else:
+            out = fwd_fn(stage_in)
         fwd_idx += 1
         return out
 
@@ -136,8 +153,13 @@ def forward_backward(fwd_fn: Callable,
             output_store.append(out)
 
     # ---------------- steady 1F1B
-    for i in range(num_steady):
-        stage_in = recv_act()
+    fused_next_in: Optional[torch.Tensor] = None  # act already received by
+    for i in range(num_steady):                   # a fused send-bwd+recv-fwd
+        if fused_next_in is not None:
+            stage_in = fused_next_in
+            fused_next_in = None
+        else:
+            stage_in = recv_act()
         out = run_forward(stage_in)
         if forward_only:
             send_act(out, first_send=(num_warmup == 0 and i == 0))
@@ -164,8 +186,17 @@ def forward_backward(fwd_fn: Callable,
         if not is_first:
             g = b_in.grad
             assert g is not None, "no grad flowed to stage input"
-            # fused send-bwd + recv-fwd happens at next loop's recv; send now
-            p2p.send_backward(g)
+            if i < num_steady - 1:
+                # fused send-bwd + recv-fwd: ONE batched p2p carries the
+                # input grad back and the next micro-batch's activation in
+                # (pairs with prev stage's send_forward_recv_backward; the
+                # reference fuses this too, comm.py:469 — round 1 issued
+                # two blocking calls here)
+                fused_next_in = p2p.send_backward_recv_forward(
+                    g, act_shape, act_dtype)
+                fused_next_in.requires_grad_(True)
+            else:
+                p2p.send_backward(g)
 
     # ---------------- cooldown backwards
     if not forward_only:
