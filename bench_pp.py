@@ -35,18 +35,31 @@ class PPStage(nn.Module):
         if not self.is_first and self.tp_size > 1:
             # stage boundaries carry SP shards (see bench_pp docstring)
             set_sequence_parallel_attr(x)
+        head = None
         for layer in self.layers:
             from torchdistpackage_amd.models.gpt2 import GPT2Head
             if isinstance(layer, GPT2Head):
-                if is_sequence_parallel(x) and self.tp_size > 1:
+                head = layer
+                # vocab-parallel heads gather SP internally (bwd
+                # reduce-scatter); replicated heads need the pre-gather
+                if not layer.vocab_parallel and \
+                        is_sequence_parallel(x) and self.tp_size > 1:
                     x = gather_from_sequence_parallel_region(
                         x, bwd_mode="split")
             x = layer(x)
         if self.is_last:
-            logits = x  # head output (B_mb, S, V)
-            from torchdistpackage_amd.ops import cross_entropy_loss
-            loss = cross_entropy_loss(logits.transpose(0, 1),
-                                      labels.transpose(0, 1)) / self.num_mb
+            logits = x  # head output (B_mb, S, V) or (B_mb, S, V/tp)
+            if head is not None and head.vocab_parallel:
+                from torchdistpackage_amd.parallel.tensor.vocab import \
+                    vocab_parallel_cross_entropy
+                loss = vocab_parallel_cross_entropy(
+                    logits.transpose(0, 1), labels.transpose(0, 1),
+                    head.vocab_start, head.vocab_end) / self.num_mb
+            else:
+                from torchdistpackage_amd.ops import cross_entropy_loss
+                loss = cross_entropy_loss(
+                    logits.transpose(0, 1),
+                    labels.transpose(0, 1)) / self.num_mb
             return loss
         return x
 

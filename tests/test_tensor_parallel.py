@@ -156,14 +156,18 @@ def test_tp_sp_transformer():
     run_distributed(_tp_sp_transformer, world_size=2)
 
 
-def _tp_gpt2(rank, world_size):
-    """GPT2Model at tp=2 (SP on) vs tp=1 full model: loss parity."""
+def _tp_gpt2(rank, world_size, vocab_parallel=True):
+    """GPT2Model at tp=2 (SP on) vs tp=1 full model: loss + grad parity.
+    With ``vocab_parallel`` the embedding/head/CE are vocab-sharded
+    (VERDICT r01 missing #5) and the wte shard grads are compared against
+    the matching slice of the oracle's full-table grads."""
     import torch.distributed as dist
     from torchdistpackage_amd.dist.topo import tpc
     from torchdistpackage_amd.parallel.tensor import set_tp_group
     from torchdistpackage_amd.models.gpt2 import GPT2Config, GPT2Model
 
-    cfg = GPT2Config(vocab_size=128, n_layer=2, n_head=4, dim=32, max_seq=16)
+    cfg = GPT2Config(vocab_size=128, n_layer=2, n_head=4, dim=32, max_seq=16,
+                     vocab_parallel=vocab_parallel)
     torch.manual_seed(0)
     # oracle runs BEFORE TP groups exist: ParallelBlock layers check the TP
     # world dynamically, so a "full" model forwarded after TP init would
@@ -179,7 +183,11 @@ def _tp_gpt2(rank, world_size):
     torch.manual_seed(0)
     tp = GPT2Model(cfg)
     # surgery: copy embeddings/head; split blocks
-    tp.embed.load_state_dict(full.embed.state_dict())
+    if vocab_parallel:
+        tp.embed.wte.load_from_full(full.embed.wte.weight)
+        tp.embed.wpe.load_state_dict(full.embed.wpe.state_dict())
+    else:
+        tp.embed.load_state_dict(full.embed.state_dict())
     tp.head.ln_f.load_state_dict(full.head.ln_f.state_dict())
     for fb, tb in zip(full.blocks, tp.blocks):
         tb.init_from_full(fb)
@@ -191,14 +199,28 @@ def _tp_gpt2(rank, world_size):
     from torchdistpackage_amd.parallel.tensor import \
         allreduce_sequence_parallel_grads
     allreduce_sequence_parallel_grads(tp)
-    # embedding grads (replicated) must match
-    assert torch.allclose(tp.embed.wte.weight.grad,
-                          full.embed.wte.weight.grad, atol=1e-4)
+    if vocab_parallel:
+        vs, ve = tp.head.vocab_start, tp.head.vocab_end
+        assert torch.allclose(tp.embed.wte.weight.grad,
+                              full.embed.wte.weight.grad[vs:ve], atol=1e-4)
+        assert torch.allclose(tp.embed.wpe.weight.grad,
+                              full.embed.wpe.weight.grad, atol=1e-4)
+        assert torch.allclose(tp.head.ln_f.weight.grad,
+                              full.head.ln_f.weight.grad, atol=1e-4)
+    else:
+        # embedding grads (replicated) must match
+        assert torch.allclose(tp.embed.wte.weight.grad,
+                              full.embed.wte.weight.grad, atol=1e-4)
     return True
 
 
 def test_tp_gpt2_model():
     run_distributed(_tp_gpt2, world_size=2)
+
+
+def test_tp_gpt2_model_replicated_head():
+    run_distributed(_tp_gpt2, world_size=2,
+                    kwargs={"vocab_parallel": False})
 
 
 def _tp_llama(rank, world_size):
@@ -222,10 +244,11 @@ def _tp_llama(rank, world_size):
     torch.manual_seed(0)
     tp = LlamaModel(cfg)
     # weight surgery: col shards rows, row shards cols
-    tp.embed.load_state_dict(full.embed.state_dict())
+    tp.embed.tok.load_from_full(full.embed.tok.weight)
     tp.head.norm.load_state_dict(full.head.norm.state_dict())
     with torch.no_grad():
-        tp.head.weight.copy_(full.head.weight)
+        tp.head.weight.copy_(
+            full.head.weight[tp.head.vocab_start:tp.head.vocab_end])
         for fb, tb in zip(full.blocks, tp.blocks):
             tb.attn_norm.load_state_dict(fb.attn_norm.state_dict())
             tb.mlp_norm.load_state_dict(fb.mlp_norm.state_dict())
@@ -247,8 +270,77 @@ def _tp_llama(rank, world_size):
     for fb, tb in zip(full.blocks, tp.blocks):
         assert torch.allclose(tb.attn_norm.weight.grad,
                               fb.attn_norm.weight.grad, atol=1e-4)
+    vs, ve = tp.head.vocab_start, tp.head.vocab_end
+    assert torch.allclose(tp.embed.tok.weight.grad,
+                          full.embed.tok.weight.grad[vs:ve], atol=1e-4)
+    assert torch.allclose(tp.head.weight.grad,
+                          full.head.weight.grad[vs:ve], atol=1e-4)
     return True
 
 
 def test_tp_llama_model():
     run_distributed(_tp_llama, world_size=2)
+
+
+def _vocab_parallel_ce(rank, world_size):
+    """vocab_parallel_cross_entropy over sharded logits vs F.cross_entropy
+    on the full logits: loss and d(logits) parity."""
+    import torch.nn.functional as F
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.tensor import set_tp_group
+    from torchdistpackage_amd.parallel.tensor.vocab import (
+        vocab_parallel_cross_entropy)
+
+    tpc.setup_process_groups([("tensor", world_size)])
+    set_tp_group(tpc.get_group("tensor"))
+
+    torch.manual_seed(3)
+    N, V = 24, 64
+    full = torch.randn(N, V)
+    target = torch.randint(0, V, (N,))
+    Vp = V // world_size
+    vs, ve = rank * Vp, (rank + 1) * Vp
+    local = full[:, vs:ve].clone().requires_grad_(True)
+    loss = vocab_parallel_cross_entropy(local, target, vs, ve)
+
+    ref = full.clone().requires_grad_(True)
+    ref_loss = F.cross_entropy(ref, target)
+    assert torch.allclose(loss, ref_loss, atol=1e-5), \
+        (float(loss), float(ref_loss))
+    loss.backward()
+    ref_loss.backward()
+    assert torch.allclose(local.grad, ref.grad[:, vs:ve], atol=1e-5)
+    return True
+
+
+def test_vocab_parallel_cross_entropy():
+    run_distributed(_vocab_parallel_ce, world_size=2)
+
+
+def _vocab_parallel_embedding(rank, world_size):
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.tensor import set_tp_group
+    from torchdistpackage_amd.parallel.tensor.vocab import (
+        VocabParallelEmbedding)
+
+    tpc.setup_process_groups([("tensor", world_size)])
+    set_tp_group(tpc.get_group("tensor"))
+    torch.manual_seed(4)
+    full_w = torch.randn(32, 8)
+    emb = VocabParallelEmbedding(32, 8)
+    emb.load_from_full(full_w)
+    idx = torch.randint(0, 32, (3, 6))
+    out = emb(idx)
+    ref = torch.nn.functional.embedding(idx, full_w)
+    assert torch.allclose(out, ref, atol=1e-5)
+    # SP output path: (S/tp, B, D) shard of the seq-first full
+    out_sp = emb(idx, sequence_parallel_out=True)
+    ref_sp = ref.transpose(0, 1)
+    shard = 6 // world_size
+    assert torch.allclose(out_sp, ref_sp[rank * shard:(rank + 1) * shard],
+                          atol=1e-5)
+    return True
+
+
+def test_vocab_parallel_embedding():
+    run_distributed(_vocab_parallel_embedding, world_size=2)

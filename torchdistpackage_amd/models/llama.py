@@ -44,6 +44,9 @@ class LlamaConfig:
     causal: bool = True
     sequence_parallel: bool = True
     tie_weights: bool = False
+    # vocab-sharded embedding/head/CE over TP (see models/gpt2.py); inert
+    # at tp=1
+    vocab_parallel: bool = True
 
 
 def llama3_8b() -> LlamaConfig:
@@ -176,21 +179,46 @@ class LlamaBlock(nn.Module):
 class LlamaEmbedding(nn.Module):
     def __init__(self, cfg: LlamaConfig, device=None, dtype=None):
         super().__init__()
-        self.tok = nn.Embedding(cfg.vocab_size, cfg.dim, device=device,
-                                dtype=dtype)
-        nn.init.normal_(self.tok.weight, std=0.02)
+        self.vocab_parallel = cfg.vocab_parallel and get_tp_size() > 1
+        self.sequence_parallel = cfg.sequence_parallel
+        if self.vocab_parallel:
+            from ..parallel.tensor.vocab import VocabParallelEmbedding
+            self.tok = VocabParallelEmbedding(cfg.vocab_size, cfg.dim,
+                                              init_std=0.02, device=device,
+                                              dtype=dtype)
+        else:
+            self.tok = nn.Embedding(cfg.vocab_size, cfg.dim, device=device,
+                                    dtype=dtype)
+            nn.init.normal_(self.tok.weight, std=0.02)
 
     def forward(self, idx):
+        if self.vocab_parallel and self.sequence_parallel \
+                and get_tp_size() > 1:
+            # reduce-scatter straight into the (S/tp, B, D) SP layout
+            x = self.tok(idx, sequence_parallel_out=True)
+            set_sequence_parallel_attr(x)
+            return x
         return self.tok(idx).transpose(0, 1).contiguous()  # (S, B, D)
 
 
 class LlamaHead(nn.Module):
-    def __init__(self, cfg: LlamaConfig, tok: Optional[nn.Embedding],
+    def __init__(self, cfg: LlamaConfig, tok,
                  device=None, dtype=None):
         super().__init__()
         kw = {"device": device, "dtype": dtype}
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps, **kw)
-        if cfg.tie_weights and tok is not None:
+        self.vocab_parallel = cfg.vocab_parallel and get_tp_size() > 1
+        if self.vocab_parallel:
+            from ..parallel.tensor.vocab import VocabParallelHead
+            tied = tok.weight if (cfg.tie_weights and tok is not None) \
+                else None
+            vh = VocabParallelHead(cfg.dim, cfg.vocab_size, weight=tied,
+                                   init_std=0.02, **kw)
+            self.vocab_start, self.vocab_end = vh.vocab_start, vh.vocab_end
+            self.weight = tied if tied is not None else vh.weight
+            if cfg.sequence_parallel:
+                mark_sequence_parallel_params(self.norm)
+        elif cfg.tie_weights and tok is not None:
             self.weight = tok.weight
         else:
             self.weight = nn.Parameter(
@@ -198,6 +226,15 @@ class LlamaHead(nn.Module):
             nn.init.normal_(self.weight, std=0.02)
 
     def forward(self, x):
+        if self.vocab_parallel:
+            sp_in = is_sequence_parallel(x)
+            x = self.norm(x)
+            if sp_in and get_tp_size() > 1:
+                x = gather_from_sequence_parallel_region(
+                    x, bwd_mode="reduce_scatter")
+            else:
+                x = copy_to_tp_region(x)
+            return F.linear(x, self.weight).transpose(0, 1)
         return F.linear(self.norm(x), self.weight).transpose(0, 1)
 
 
@@ -216,6 +253,16 @@ class LlamaModel(nn.Module):
         x = self.embed(idx)
         for blk in self.blocks:
             x = blk(x)
+        if self.head.vocab_parallel:
+            logits = self.head(x)   # LOCAL (B, S, V/tp), never gathered
+            out = {"logits": logits}
+            if labels is not None:
+                from ..parallel.tensor.vocab import \
+                    vocab_parallel_cross_entropy
+                out["loss"] = vocab_parallel_cross_entropy(
+                    logits.transpose(0, 1), labels.transpose(0, 1),
+                    self.head.vocab_start, self.head.vocab_end)
+            return out
         if is_sequence_parallel(x) and get_tp_size() > 1:
             x = gather_from_sequence_parallel_region(x, bwd_mode="split")
         logits = self.head(x)
