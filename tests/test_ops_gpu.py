@@ -403,3 +403,74 @@ def test_tr16_transpose_read_semantics():
                 src = g0 + ((l >> 2) & 3) + 4 * j
                 expect = addr_bytes(src, mode) // 8 * 4 + (l & 3)
                 assert r[l][j] == expect, (mode, l, j, r[l].tolist(), expect)
+
+
+# ------------------------------------------------------------- custom GEMM
+
+def _gemm_yardstick(out, ref, lib_out):
+    """err(mine) must be comparable to err(hipBLASLt) vs the fp32 oracle."""
+    e_mine = ((out.float() - ref).abs().max() / ref.abs().max()).item()
+    e_lib = ((lib_out.float() - ref).abs().max() / ref.abs().max()).item()
+    assert e_mine <= max(2.5 * e_lib, 1e-3), (e_mine, e_lib)
+
+
+@pytest.mark.parametrize("M,N,K", [(512, 512, 512), (256, 256, 64),
+                                   (512, 256, 160)])
+def test_gemm_fprop(M, N, K):
+    torch.manual_seed(0)
+    x = (torch.randn(M, K) * 0.5).bfloat16().to(_dev())
+    w = (torch.randn(N, K) * 0.5).bfloat16().to(_dev())
+    b = torch.randn(N).bfloat16().to(_dev())
+    ref = x.float() @ w.float().t()
+    _gemm_yardstick(ops.ext("gemm").gemm_fprop(x, w, None), ref, x @ w.t())
+    _gemm_yardstick(ops.ext("gemm").gemm_fprop(x, w, b), ref + b.float(),
+                    x @ w.t() + b)
+
+
+@pytest.mark.parametrize("kswz", [False, True])
+def test_gemm_dgrad(kswz):
+    torch.manual_seed(1)
+    M, N, K = 512, 512, 768   # dy (M,K) @ w (K,N)
+    dy = (torch.randn(M, K) * 0.5).bfloat16().to(_dev())
+    w = (torch.randn(K, N) * 0.5).bfloat16().to(_dev())
+    ref = dy.float() @ w.float()
+    _gemm_yardstick(ops.ext("gemm").gemm_dgrad(dy, w, kswz), ref, dy @ w)
+
+
+@pytest.mark.parametrize("kswz", [False, True])
+@pytest.mark.parametrize("sk", [1, 2, 4])
+def test_gemm_wgrad(kswz, sk):
+    torch.manual_seed(2)
+    T, M, N = 2048, 512, 256
+    dy = (torch.randn(T, M) * 0.5).bfloat16().to(_dev())
+    x = (torch.randn(T, N) * 0.5).bfloat16().to(_dev())
+    ref = dy.float().t() @ x.float()
+    _gemm_yardstick(ops.ext("gemm").gemm_wgrad(dy, x, sk, kswz), ref,
+                    dy.t() @ x)
+
+
+def test_gemm_autograd_linear():
+    """linear() dispatch end-to-end vs F.linear autograd (fp32 oracle)."""
+    from torchdistpackage_amd.ops.gemm import linear as fast_linear
+    torch.manual_seed(3)
+    M, N, K = 512, 512, 512
+    x0 = (torch.randn(M, K) * 0.5).bfloat16().to(_dev())
+    w0 = (torch.randn(N, K) * 0.02).bfloat16().to(_dev())
+    b0 = torch.randn(N).bfloat16().to(_dev())
+
+    x1 = x0.clone().requires_grad_(True)
+    w1 = w0.clone().requires_grad_(True)
+    b1 = b0.clone().requires_grad_(True)
+    out1 = fast_linear(x1, w1, b1)
+    out1.float().pow(2).mean().backward()
+
+    x2 = x0.clone().float().requires_grad_(True)
+    w2 = w0.clone().float().requires_grad_(True)
+    b2 = b0.clone().float().requires_grad_(True)
+    out2 = torch.nn.functional.linear(x2, w2, b2)
+    out2.pow(2).mean().backward()
+
+    for g1, g2 in [(out1, out2), (x1.grad, x2.grad), (w1.grad, w2.grad),
+                   (b1.grad, b2.grad)]:
+        e = (g1.float() - g2).abs().max() / g2.abs().max().clamp_min(1e-6)
+        assert e.item() < 0.06, e.item()

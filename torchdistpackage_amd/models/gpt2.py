@@ -21,6 +21,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import LayerNorm
+from ..ops.gemm import linear as fast_linear
 from ..parallel.tensor import ParallelBlock, get_tp_size
 
 
@@ -147,10 +148,10 @@ class GPT2Head(nn.Module):
                 # replicated activations: dx partial per rank -> bwd
                 # all-reduce via the copy region
                 x = copy_to_tp_region(x)
-            logits = F.linear(x, self.weight)
+            logits = fast_linear(x, self.weight)
             return logits.transpose(0, 1)
         x = self.ln_f(x)
-        logits = F.linear(x, self.weight)
+        logits = fast_linear(x, self.weight)
         return logits.transpose(0, 1)
 
 

@@ -21,6 +21,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops import RMSNorm, flash_attention
+from ..ops.gemm import linear as fast_linear
 from ..parallel.tensor.tp_utils import (ColParallelLinear, RowParallelLinear,
                                         copy_to_tp_region,
                                         gather_from_sequence_parallel_region,
@@ -109,9 +110,9 @@ class LlamaAttention(nn.Module):
         else:
             x = copy_to_tp_region(x)
         S, B, _ = x.shape
-        q = F.linear(x, self.wq.weight)
-        k = F.linear(x, self.wk.weight)
-        v = F.linear(x, self.wv.weight)
+        q = fast_linear(x, self.wq.weight)
+        k = fast_linear(x, self.wk.weight)
+        v = fast_linear(x, self.wv.weight)
 
         def shape(t, nh):
             return t.reshape(S, B, nh, self.hd).permute(1, 2, 0, 3).contiguous()
@@ -143,8 +144,8 @@ class LlamaMlp(nn.Module):
             x = gather_from_sequence_parallel_region(x)
         else:
             x = copy_to_tp_region(x)
-        return self.w2(F.silu(F.linear(x, self.w1.weight)) *
-                       F.linear(x, self.w3.weight))
+        return self.w2(F.silu(fast_linear(x, self.w1.weight)) *
+                       fast_linear(x, self.w3.weight))
 
 
 class LlamaBlock(nn.Module):
@@ -234,8 +235,8 @@ class LlamaHead(nn.Module):
                     x, bwd_mode="reduce_scatter")
             else:
                 x = copy_to_tp_region(x)
-            return F.linear(x, self.weight).transpose(0, 1)
-        return F.linear(self.norm(x), self.weight).transpose(0, 1)
+            return fast_linear(x, self.weight).transpose(0, 1)
+        return fast_linear(self.norm(x), self.weight).transpose(0, 1)
 
 
 class LlamaModel(nn.Module):

@@ -83,9 +83,10 @@ class Expert(nn.Module):
         nn.init.normal_(self.fc2.weight, std=0.02)
 
     def forward(self, x):
-        h = F.linear(x, self.fc1.weight)
+        from ..ops.gemm import linear as fast_linear
+        h = fast_linear(x, self.fc1.weight)
         h = bias_gelu(h, self.fc1.bias)
-        return self.fc2(h)
+        return fast_linear(h, self.fc2.weight, self.fc2.bias)
 
 
 def _all_to_all_uneven(x: torch.Tensor, in_splits: List[int],

@@ -24,6 +24,7 @@ SOURCES = [
     "attention_v2.hip",
     "cross_entropy.hip",
     "probe.hip",
+    "gemm.hip",
 ]
 
 

@@ -39,6 +39,11 @@ torch::Tensor mfma_probe_16x16x32(torch::Tensor a, torch::Tensor b);
 torch::Tensor mfma_probe_32x32x16(torch::Tensor a, torch::Tensor b);
 torch::Tensor tr16_probe(long addr_mode);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
+torch::Tensor gemm_fprop(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias);
+torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w, bool kswz);
+torch::Tensor gemm_wgrad(torch::Tensor dy, torch::Tensor x, long splitk,
+                         bool kswz);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);
 
@@ -60,5 +65,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe_32x32x16", &mfma_probe_32x32x16);
   m.def("tr16_probe", &tr16_probe);
   m.def("ce_fwd", &ce_fwd);
+  m.def("gemm_fprop", &gemm_fprop);
+  m.def("gemm_dgrad", &gemm_dgrad);
+  m.def("gemm_wgrad", &gemm_wgrad);
   m.def("ce_bwd", &ce_bwd);
 }

@@ -1,0 +1,348 @@
+// Hand-written CDNA4 bf16 GEMM for the transformer hot shapes (gfx950).
+//
+// Replaces hipBLASLt on the model GEMMs the reference delegates to cuBLAS
+// (/root/reference/torchdistpackage/parallel/tensor_parallel/tp_utils.py:171
+// torch.matmul): QKV / out-proj / MLP fprop, dgrad, and the K=16384 wgrad
+// family that hipBLASLt runs at ~1.03 PF/s (profiles/r01_notes.md item 4).
+//
+// Structure = the guide's 256x256 8-phase template (cdna_hip_programming.md
+// §5): BM=BN=256, 8 waves (2Mx4N, 128x64 per wave), K staged in 32-deep
+// slots through a 4-slot LDS ring per operand (128 KiB total, ONE __shared__
+// object), global->LDS via global_load_lds dwordx4 (2 per thread per phase),
+// prefetch 3 slots ahead with counted s_waitcnt vmcnt(8) at slot boundaries
+// (never vmcnt(0) in the main loop), raw s_barrier phase boundaries, and
+// s_setprio(1) around each 16-MFMA cluster (v_mfma_f32_16x16x32_bf16).
+//
+// Operand layouts (element (row, k)):
+//   LAY=0 "k-inner": row-major (rows, K)  — fprop x and nn.Linear weight.
+//     LDS slot image [256 rows][32 k] (64B rows), 16B-chunk XOR swizzle
+//     c^((row>>2)&1)<<1 applied on the glds SOURCE address (rule 21) so
+//     ds_read_b128 A/B-fragment reads are bank-conflict-free.
+//   LAY=1 "k-outer": row-major (K, rows) — wgrad operands (dY, X viewed
+//     along tokens) and dgrad's weight.  LDS slot image [32 k][256 rows]
+//     (512B rows, fully-coalesced glds); fragments gathered with
+//     ds_read_b64_tr_b16 pairs using the probe-verified lane model
+//     (probe.hip tr16_probe; same recipe as attention_v2.hip dq-lite),
+//     optional 16B-chunk swizzle c^((k&3)<<2) against tr-read conflicts.
+//
+// Split-K (wgrad M=N=2048-class shapes: too few 256^2 tiles to fill 256
+// CUs): blockIdx carries tile*splitk+slice after the bijective XCD remap
+// (T1) so a tile's slices share an XCD; slices write fp32 slabs, a separate
+// reduce kernel folds them to bf16 (slab is MBytes-scale, far past the
+// guide's in-launch-combine regime).
+//
+// MFMA fragment maps (probe-verified on hardware, see probe.hip):
+//   A: lane l holds A[row=l&15][k=(l>>4)*8+j], j=0..7
+//   B: lane l holds B[col=l&15][k=(l>>4)*8+j]   (B stored (N,K): same map)
+//   C/D: col=l&15, row=(l>>4)*4+reg
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int SLOT_HW = 256 * 32;     // one staged slot: 8192 hw = 16 KiB
+constexpr int BREG_HW = 4 * SLOT_HW;  // B operand region (halfwords)
+
+#define GLDS16(gp, lp)                                              \
+  __builtin_amdgcn_global_load_lds(                                 \
+      (const __attribute__((address_space(1))) unsigned int*)(gp),  \
+      (__attribute__((address_space(3))) unsigned int*)(lp), 16, 0, 0)
+
+DEVINL unsigned lds_addr_of(const ushort* p) {
+  return (unsigned)(unsigned long long)(
+      __attribute__((address_space(3))) const ushort*)p;
+}
+
+// bijective XCD-aware remap (T1): each XCD gets a contiguous chunk of the
+// logical grid so neighbouring tiles (and a tile's split-K slices) share L2
+DEVINL int xcd_remap(int wg, int nwg) {
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = wg & 7, idx = wg >> 3;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+template <int LAYA, int LAYB, bool SPLIT, bool KSWZ>
+__launch_bounds__(512)
+__global__ void gemm256_kernel(const ushort* __restrict__ A,
+                               const ushort* __restrict__ B,
+                               const ushort* __restrict__ bias,
+                               ushort* __restrict__ C,
+                               float* __restrict__ slab,
+                               int M, int N, int K,
+                               long lda, long ldb, long ldc,
+                               int splitk, int kper, int tiles_n) {
+  __shared__ __attribute__((aligned(16))) ushort lds[8 * SLOT_HW];  // 128 KiB
+
+  const int nwg = gridDim.x;
+  const int wgid = xcd_remap(blockIdx.x, nwg);
+  const int tile = wgid / splitk;
+  const int slice = wgid - tile * splitk;
+  const int tm = tile / tiles_n;
+  const int tn = tile - tm * tiles_n;
+
+  const int tid = threadIdx.x;
+  const int w = tid >> 6;            // wave id (wave-uniform)
+  const int lane = tid & 63;
+  const int wm = w >> 2, wn = w & 3; // wave tile (wm*128, wn*64)
+  const int l15 = lane & 15, lg = lane >> 4;
+
+  const ushort* Abase = (LAYA == 0) ? A + (long)(tm * 256) * lda
+                                    : A + tm * 256;
+  const ushort* Bbase = (LAYB == 0) ? B + (long)(tn * 256) * ldb
+                                    : B + tn * 256;
+  const int k0 = slice * kper;
+  const int nslot = kper / 32;
+
+  // ---- staging: 2 glds per thread per call (one 16 KiB slot)
+  auto stage = [&](const ushort* gb, long ld, int s, int region, int lay) {
+    const int ss = s < nslot ? s : nslot - 1;  // tail clamp: harmless
+    const int kk = k0 + ss * 32;               // re-stage, never read
+    const int ring = s & 3;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int dhw = region + ring * SLOT_HW + w * 512 + i * 4096;
+      const ushort* g;
+      if (lay == 0) {
+        const int o = w * 1024 + i * 8192 + lane * 16;  // byte within slot
+        const int row = o >> 6;                          // 64 B rows
+        const int cs = (lane & 3) ^ (((row >> 2) & 1) << 1);
+        g = gb + (long)row * ld + kk + cs * 8;
+      } else {
+        const int o = w * 1024 + i * 8192 + lane * 16;
+        const int krow = o >> 9;                         // 512 B rows
+        const int c = lane & 31;
+        const int cs = KSWZ ? (c ^ ((krow & 3) << 2)) : c;
+        g = gb + (long)(kk + krow) * ld + cs * 8;
+      }
+      GLDS16(g, lds + dhw);
+    }
+  };
+  auto stageA = [&](int s) { stage(Abase, lda, s, 0, LAYA); };
+  auto stageB = [&](int s) { stage(Bbase, ldb, s, BREG_HW, LAYB); };
+
+  // ---- fragment reads
+  // k-inner: ds_read_b128 at swizzled per-lane base + frag*1024B immediates
+  const int csr = lg ^ (((l15 >> 2) & 1) << 1);
+  const int rdA = (wm * 128 + l15) * 32 + csr * 8;             // hw offset
+  const int rdB = BREG_HW + (wn * 64 + l15) * 32 + csr * 8;
+  // k-outer: tr16 per-lane byte base (probe model); frag adds rowbase*2
+  // (XOR'd with j<<6 when swizzled)
+  const unsigned trlane = (unsigned)((8 * lg + ((lane >> 2) & 3)) * 512 +
+                                     ((lane & 3) >> 1) * 16 + (lane & 1) * 8);
+  const unsigned trJ = (unsigned)(((lane >> 2) & 3) << 6);
+  const unsigned lds0 = lds_addr_of(lds);
+
+  auto tr16pair = [&](unsigned addr) -> bf16x8_v {
+    unsigned long long lo, hi;
+    asm volatile(
+        "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+        "ds_read_b64_tr_b16 %1, %2 offset:2048\n\t"
+        "s_waitcnt lgkmcnt(0)"
+        : "=v"(lo), "=v"(hi)
+        : "v"(addr));
+    bf16x8_v r;
+    ((unsigned long long*)&r)[0] = lo;
+    ((unsigned long long*)&r)[1] = hi;
+    return r;
+  };
+  auto readA = [&](int ring, int mf) -> bf16x8_v {
+    if (LAYA == 0)
+      return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdA + mf * 512];
+    const unsigned r16 = (unsigned)((wm * 128 + mf * 16) * 2);
+    const unsigned a = lds0 + (unsigned)(ring * SLOT_HW * 2) + trlane +
+                       (KSWZ ? (r16 ^ trJ) : r16);
+    return tr16pair(a);
+  };
+  auto readB = [&](int ring, int nf) -> bf16x8_v {
+    if (LAYB == 0)
+      return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdB + nf * 512];
+    const unsigned r16 = (unsigned)((wn * 64 + nf * 16) * 2);
+    const unsigned a = lds0 + (unsigned)(BREG_HW * 2) +
+                       (unsigned)(ring * SLOT_HW * 2) + trlane +
+                       (KSWZ ? (r16 ^ trJ) : r16);
+    return tr16pair(a);
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = (f32x4)(0.f);
+
+  // ---- prologue: slots 0..2 staged; wait until slot 0 landed
+  stageA(0); stageB(0);
+  stageA(1); stageB(1);
+  stageA(2); stageB(2);
+  asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+
+  // ---- main loop: 2 phases per 32-deep slot
+  for (int s = 0; s < nslot; ++s) {
+    const int ring = s & 3;
+    // phase a: stage A(s+3) | read A frags + B frags 0-1 | 16 MFMA
+    stageA(s + 3);
+    bf16x8_v af[8], bf0, bf1;
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) af[mf] = readA(ring, mf);
+    bf0 = readB(ring, 0);
+    bf1 = readB(ring, 1);
+    asm volatile("s_barrier" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) {
+      acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bf0, acc[mf][0], 0, 0, 0);
+      acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bf1, acc[mf][1], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_barrier" ::: "memory");
+    // phase b: stage B(s+3) | read B frags 2-3 | 16 MFMA
+    stageB(s + 3);
+    bf0 = readB(ring, 2);
+    bf1 = readB(ring, 3);
+    asm volatile("s_barrier" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) {
+      acc[mf][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bf0, acc[mf][2], 0, 0, 0);
+      acc[mf][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bf1, acc[mf][3], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    // slot boundary: slot s+1 must have landed (counted wait, never 0)
+    asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+  }
+
+  // ---- epilogue
+  const long crow0 = (long)tm * 256 + wm * 128;
+  const int ccol0 = tn * 256 + wn * 64;
+  if (SPLIT) {
+    float* sl = slab + (long)slice * M * N;
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const long row = crow0 + mf * 16 + lg * 4;
+        const int col = ccol0 + nf * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          sl[(row + r) * (long)N + col] = acc[mf][nf][r];
+      }
+  } else {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int col = ccol0 + nf * 16 + l15;
+      const float bv = bias ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf) {
+        const long row = crow0 + mf * 16 + lg * 4;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          C[(row + r) * ldc + col] = f2bf(acc[mf][nf][r] + bv);
+      }
+    }
+  }
+}
+
+// fold SPLITK fp32 slabs into bf16 (grid-strided, float4-vectorized)
+__global__ void splitk_reduce_kernel(const float* __restrict__ slab,
+                                     ushort* __restrict__ C, long total,
+                                     int splitk) {
+  const long i4 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (i4 >= total) return;
+  float4 acc = *(const float4*)(slab + i4);
+  for (int s = 1; s < splitk; ++s) {
+    const float4 v = *(const float4*)(slab + (long)s * total + i4);
+    acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+  }
+  ushort4 o;
+  o.x = f2bf(acc.x); o.y = f2bf(acc.y); o.z = f2bf(acc.z); o.w = f2bf(acc.w);
+  *(ushort4*)(C + i4) = o;
+}
+
+void check_bf16_2d(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
+              t.dim() == 2 && t.is_contiguous(), name,
+              " must be contiguous 2-D bf16 CUDA");
+}
+
+template <int LAYA, int LAYB>
+torch::Tensor launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
+                          const c10::optional<torch::Tensor>& bias,
+                          int M, int N, int K, long lda, long ldb,
+                          int splitk, bool kswz) {
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0,
+              "tdpa_gemm: M,N must be multiples of 256 (got ", M, "x", N,
+              ") — dispatch layer should fall back");
+  TORCH_CHECK(splitk >= 1 && K % (32 * splitk) == 0,
+              "tdpa_gemm: K (", K, ") must be a multiple of 32*splitk");
+  auto C = torch::empty({M, N}, A.options());
+  const int tiles_m = M / 256, tiles_n = N / 256;
+  const int nwg = tiles_m * tiles_n * splitk;
+  const ushort* bptr = nullptr;
+  if (bias.has_value()) {
+    TORCH_CHECK(bias->numel() == N && bias->scalar_type() == torch::kBFloat16);
+    bptr = (const ushort*)bias->data_ptr();
+  }
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (splitk == 1) {
+    auto kern = kswz ? gemm256_kernel<LAYA, LAYB, false, true>
+                     : gemm256_kernel<LAYA, LAYB, false, false>;
+    hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 0, stream,
+                       (const ushort*)A.data_ptr(), (const ushort*)B.data_ptr(),
+                       bptr, (ushort*)C.data_ptr(), nullptr, M, N, K,
+                       lda, ldb, (long)N, 1, K, tiles_n);
+  } else {
+    auto slab = torch::empty({splitk, (long)M * N},
+                             A.options().dtype(torch::kFloat));
+    auto kern = kswz ? gemm256_kernel<LAYA, LAYB, true, true>
+                     : gemm256_kernel<LAYA, LAYB, true, false>;
+    hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 0, stream,
+                       (const ushort*)A.data_ptr(), (const ushort*)B.data_ptr(),
+                       nullptr, nullptr, slab.data_ptr<float>(), M, N, K,
+                       lda, ldb, (long)N, splitk, K / splitk, tiles_n);
+    const long total = (long)M * N;
+    const long nb = (total / 4 + 255) / 256;
+    hipLaunchKernelGGL(splitk_reduce_kernel, dim3(nb), dim3(256), 0, stream,
+                       slab.data_ptr<float>(), (ushort*)C.data_ptr(), total,
+                       splitk);
+  }
+  HIP_CHECK_LAST();
+  return C;
+}
+
+}  // namespace
+
+// y (M,N) = x (M,K) @ w(N,K)^T + bias — nn.Linear forward.
+torch::Tensor gemm_fprop(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias) {
+  check_bf16_2d(x, "x"); check_bf16_2d(w, "w");
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch");
+  return launch_gemm<0, 0>(x, w, bias, M, N, K, K, K, 1, false);
+}
+
+// dx (M,Kin) = dy (M,Kout) @ w (Kout,Kin) — nn.Linear input grad.
+torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w, bool kswz) {
+  check_bf16_2d(dy, "dy"); check_bf16_2d(w, "w");
+  const int M = dy.size(0), K = dy.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(0) == K, "shape mismatch");
+  return launch_gemm<0, 1>(dy, w, c10::nullopt, M, N, K, K, N, 1, kswz);
+}
+
+// dw (Dout,Din) = dy (T,Dout)^T @ x (T,Din) — nn.Linear weight grad,
+// K = tokens (the measured-slow hipBLASLt family at K=16384).
+torch::Tensor gemm_wgrad(torch::Tensor dy, torch::Tensor x, long splitk,
+                         bool kswz) {
+  check_bf16_2d(dy, "dy"); check_bf16_2d(x, "x");
+  const int K = dy.size(0), M = dy.size(1), N = x.size(1);
+  TORCH_CHECK(x.size(0) == K, "shape mismatch");
+  return launch_gemm<1, 1>(dy, x, c10::nullopt, M, N, K, M, N,
+                           (int)splitk, kswz);
+}

@@ -1,0 +1,98 @@
+"""Dispatch layer for the hand-written CDNA4 GEMM (csrc/gemm.hip).
+
+``linear(x, w, bias)`` is a drop-in for ``F.linear`` on the transformer hot
+path: on GPU, bf16, and 256-aligned shapes it routes fprop/dgrad/wgrad
+through the in-tree 256x256 MFMA kernel (wgrad with a split-K heuristic
+sized for 256 CUs); anything else falls back to ``F.linear`` (hipBLASLt).
+Gate with env ``TDPA_GEMM=0`` to force the library path for A/B runs.
+
+The reference delegates these GEMMs to cuBLAS via torch.matmul
+(tp_utils.py:171); BASELINE.json's north star names QKV / out-proj as
+hand-written CDNA4 — this is that path.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import ext
+
+_ENABLED = os.environ.get("TDPA_GEMM", "1") != "0"
+# wgrad K=16384-class shapes keep fp32 partials in a slab; cap its size
+_MAX_SLAB_BYTES = 2 << 30
+
+
+def _supported_mnk(M: int, N: int, K: int) -> bool:
+    return M % 256 == 0 and N % 256 == 0 and K % 32 == 0 and K >= 32
+
+
+def pick_splitk(M: int, N: int, K: int) -> int:
+    """Fill the 256-CU chip: prefer the split factor whose grid is closest
+    to a whole multiple of 256 blocks (1 block/CU kernel)."""
+    ntiles = (M // 256) * (N // 256)
+    best, best_eff = 1, 0.0
+    for sk in (1, 2, 4, 8):
+        if K % (32 * sk) or K // sk < 128:
+            continue
+        if sk > 1 and sk * M * N * 4 > _MAX_SLAB_BYTES:
+            continue
+        blocks = ntiles * sk
+        waves = -(-blocks // 256)
+        eff = blocks / (256.0 * waves) / (1.0 + 0.03 * (sk - 1))
+        if eff > best_eff + 1e-9:
+            best, best_eff = sk, eff
+    return best
+
+
+def gemm_enabled() -> bool:
+    return _ENABLED
+
+
+class _TdpaLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2d, weight, bias):
+        ctx.save_for_backward(x2d, weight)
+        ctx.has_bias = bias is not None
+        return ext("gemm_fprop").gemm_fprop(x2d, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        M, K = x2d.shape
+        N = weight.shape[0]
+        e = ext("gemm_bwd")
+        # fprop gating ensures M%256, N%256, K%32; both backward GEMMs
+        # additionally need the in-features dim 256-aligned
+        if K % 256 == 0:
+            dx = e.gemm_dgrad(dy, weight, False)
+            dw = e.gemm_wgrad(dy, x2d, pick_splitk(N, K, M), False)
+        else:
+            dx = dy @ weight
+            dw = dy.t() @ x2d
+        db = dy.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """F.linear drop-in; routes 256-aligned bf16 GPU shapes through the
+    in-tree MFMA GEMM (fprop + both backward GEMMs)."""
+    if (_ENABLED and x.is_cuda and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and (bias is None or bias.dtype == torch.bfloat16)):
+        shape = x.shape
+        M = x.numel() // shape[-1]
+        K = shape[-1]
+        N = weight.shape[0]
+        if _supported_mnk(M, N, K):
+            x2d = x.reshape(M, K)
+            if not x2d.is_contiguous():
+                x2d = x2d.contiguous()
+            out = _TdpaLinearFn.apply(x2d, weight.contiguous(), bias)
+            return out.reshape(*shape[:-1], N)
+    return F.linear(x, weight, bias)

@@ -77,7 +77,8 @@ class TpAttention(nn.Module):
             x = gather_from_sequence_parallel_region(x)
         else:
             x = copy_to_tp_region(x)
-        qkv = F.linear(x, self.qkv.weight, self.qkv.bias)
+        from ...ops.gemm import linear as fast_linear
+        qkv = fast_linear(x, self.qkv.weight, self.qkv.bias)
         # qkv layout per rank: [q_local | k_local | v_local] thanks to the
         # interleaved loader (init_qkv_weight_from_full) / native init
         o = _sdpa(qkv, self.n_head_local, self.causal)

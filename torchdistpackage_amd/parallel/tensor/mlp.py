@@ -67,7 +67,8 @@ class TpMlp(nn.Module):
             x = gather_from_sequence_parallel_region(x)
         else:
             x = copy_to_tp_region(x)
-        h = F.linear(x, self.fc1.weight)  # bias deferred to fused kernel
+        from ...ops.gemm import linear as fast_linear
+        h = fast_linear(x, self.fc1.weight)  # bias deferred to fused kernel
         h = bias_gelu(h, self.fc1.bias)
         if self.dropout > 0 and self.training:
             h = F.dropout(h, p=self.dropout)

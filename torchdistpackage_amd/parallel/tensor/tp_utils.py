@@ -32,6 +32,8 @@ import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ...ops.gemm import linear as fast_linear
+
 _TP_GROUP: Optional[dist.ProcessGroup] = None
 
 
@@ -283,7 +285,7 @@ class TpLinear(nn.Module):
             nn.init.zeros_(self.bias)
 
     def forward(self, x):
-        return F.linear(x, self.weight, self.bias)
+        return fast_linear(x, self.weight, self.bias)
 
 
 class ColParallelLinear(TpLinear):
@@ -304,7 +306,7 @@ class ColParallelLinear(TpLinear):
 
     def forward(self, x):
         x = copy_to_tp_region(x)
-        out = F.linear(x, self.weight, self.bias)
+        out = fast_linear(x, self.weight, self.bias)
         if self.gather_output and get_tp_size() > 1:
             out = _GatherLastDim.apply(out)
         return out
@@ -354,7 +356,7 @@ class RowParallelLinear(TpLinear):
 
     def forward(self, x):
         # bias added once, after the reduction (not per-rank!)
-        out = F.linear(x, self.weight)
+        out = fast_linear(x, self.weight)
         if self.sequence_parallel and get_tp_size() > 1:
             out = reduce_scatter_to_sequence_parallel_region(out)
         else:
