@@ -325,24 +325,49 @@ def is_using_pp() -> bool:
 
 
 def test_comm(verbose: bool = False):
-    """Probe every initialized group with a small all-reduce + broadcast.
+    """Probe every initialized group with all-reduce, broadcast, all-gather
+    and ring p2p (send/recv to group neighbours).
 
-    Reference parity: process_topo.py:267-316.  Uses a (100, 128) tensor on the
-    current device.
+    Reference parity: process_topo.py:267-316 (which also probes p2p and
+    all-gather — round 1 covered only all-reduce + broadcast).  Uses a
+    (100, 128) tensor on the current device.
     """
     if not dist.is_initialized():
         raise RuntimeError("torch.distributed not initialized")
     dev = torch.device("cuda", torch.cuda.current_device()) \
         if torch.cuda.is_available() else torch.device("cpu")
+    rank = dist.get_rank()
     for mode, group in tpc._groups.items():
-        t = torch.full((100, 128), float(dist.get_rank()), device=dev)
+        ranks = tpc.get_ranks_in_group(mode)
+        t = torch.full((100, 128), float(rank), device=dev)
         dist.all_reduce(t, group=group)
-        expected = float(sum(tpc.get_ranks_in_group(mode)))
+        expected = float(sum(ranks))
         assert torch.allclose(t, torch.full_like(t, expected)), \
             f"all_reduce mismatch in group {mode}"
-        b = torch.full((8,), float(tpc.get_ranks_in_group(mode)[0]), device=dev)
-        dist.broadcast(b, src=tpc.get_ranks_in_group(mode)[0], group=group)
-        if verbose and dist.get_rank() == 0:
-            print(f"[test_comm] group '{mode}' ok "
-                  f"(ranks={tpc.get_ranks_in_group(mode)})")
+        b = torch.full((8,), float(ranks[0]), device=dev)
+        dist.broadcast(b, src=ranks[0], group=group)
+        assert torch.allclose(b, torch.full_like(b, float(ranks[0]))), \
+            f"broadcast mismatch in group {mode}"
+        # all-gather: every member's rank shows up in its slot
+        g = torch.full((4,), float(rank), device=dev)
+        out = [torch.empty_like(g) for _ in ranks]
+        dist.all_gather(out, g, group=group)
+        for slot, r in zip(out, ranks):
+            assert torch.allclose(slot, torch.full_like(slot, float(r))), \
+                f"all_gather mismatch in group {mode}"
+        # ring p2p: send to next, receive from prev (skip trivial groups)
+        if len(ranks) > 1:
+            idx = ranks.index(rank)
+            nxt = ranks[(idx + 1) % len(ranks)]
+            prv = ranks[(idx - 1) % len(ranks)]
+            s = torch.full((16,), float(rank), device=dev)
+            r = torch.empty_like(s)
+            ops = [dist.P2POp(dist.isend, s, nxt),
+                   dist.P2POp(dist.irecv, r, prv)]
+            for w in dist.batch_isend_irecv(ops):
+                w.wait()
+            assert torch.allclose(r, torch.full_like(r, float(prv))), \
+                f"p2p ring mismatch in group {mode}"
+        if verbose and rank == 0:
+            print(f"[test_comm] group '{mode}' ok (ranks={ranks})")
     dist.barrier()
