@@ -98,11 +98,12 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   const int k0 = slice * kper;
   const int nslot = kper / 32;
 
-  // ---- staging: 2 glds per thread per call (one 16 KiB slot)
-  auto stage = [&](const ushort* gb, long ld, int s, int region, int lay) {
-    const int ss = s < nslot ? s : nslot - 1;  // tail clamp: harmless
-    const int kk = k0 + ss * 32;               // re-stage, never read
-    const int ring = s & 3;
+  // ---- staging: 2 glds per thread per call (one 16 KiB slot).
+  // ring = destination ring slot; ss = SOURCE slot (clamped at the tail:
+  // the target ring slot is no longer read, the data is garbage-by-design)
+  auto stage = [&](const ushort* gb, long ld, int ring, int ss, int region,
+                   int lay) {
+    const int kk = k0 + ss * 32;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       const int dhw = region + ring * SLOT_HW + w * 512 + i * 4096;
@@ -122,8 +123,12 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
       GLDS16(g, lds + dhw);
     }
   };
-  auto stageA = [&](int s) { stage(Abase, lda, s, 0, LAYA); };
-  auto stageB = [&](int s) { stage(Bbase, ldb, s, BREG_HW, LAYB); };
+  auto stageA = [&](int s) {
+    stage(Abase, lda, s & 3, s < nslot ? s : nslot - 1, 0, LAYA);
+  };
+  auto stageB = [&](int s) {
+    stage(Bbase, ldb, s & 3, s < nslot ? s : nslot - 1, BREG_HW, LAYB);
+  };
 
   // ---- fragment reads
   // k-inner: ds_read_b128 at swizzled per-lane base + frag*1024B immediates
@@ -174,50 +179,60 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) acc[mf][nf] = (f32x4)(0.f);
 
+  // one 32-deep slot: 2 phases.  All 12 fragment reads issue at the top
+  // (counted lgkm waits let phase b's operands arrive under phase a's
+  // MFMAs); staging for slot s+3 is split across the two phases.
+  auto do_slot = [&](int s, int ring, bool stage_ok) {
+    stage(Abase, lda, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1, 0, LAYA);
+    bf16x8_v af[8], bfr[4];
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) af[mf] = readA(ring, mf);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) bfr[nf] = readB(ring, nf);
+    asm volatile("s_barrier" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) {
+      acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bfr[0], acc[mf][0], 0, 0, 0);
+      acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bfr[1], acc[mf][1], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_barrier" ::: "memory");
+    stage(Bbase, ldb, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1,
+          BREG_HW, LAYB);
+    asm volatile("s_barrier" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) {
+      acc[mf][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bfr[2], acc[mf][2], 0, 0, 0);
+      acc[mf][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          af[mf], bfr[3], acc[mf][3], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    // slot boundary: slot s+1 must have landed (counted wait, never 0)
+    asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+  };
+
   // ---- prologue: slots 0..2 staged; wait until slot 0 landed
   stageA(0); stageB(0);
   stageA(1); stageB(1);
   stageA(2); stageB(2);
   asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
 
-  // ---- main loop: 2 phases per 32-deep slot
-  for (int s = 0; s < nslot; ++s) {
-    const int ring = s & 3;
-    // phase a: stage A(s+3) | read A frags + B frags 0-1 | 16 MFMA
-    stageA(s + 3);
-    bf16x8_v af[8], bf0, bf1;
-#pragma unroll
-    for (int mf = 0; mf < 8; ++mf) af[mf] = readA(ring, mf);
-    bf0 = readB(ring, 0);
-    bf1 = readB(ring, 1);
-    asm volatile("s_barrier" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mf = 0; mf < 8; ++mf) {
-      acc[mf][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          af[mf], bf0, acc[mf][0], 0, 0, 0);
-      acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          af[mf], bf1, acc[mf][1], 0, 0, 0);
-    }
-    __builtin_amdgcn_s_setprio(0);
-    asm volatile("s_barrier" ::: "memory");
-    // phase b: stage B(s+3) | read B frags 2-3 | 16 MFMA
-    stageB(s + 3);
-    bf0 = readB(ring, 2);
-    bf1 = readB(ring, 3);
-    asm volatile("s_barrier" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mf = 0; mf < 8; ++mf) {
-      acc[mf][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          af[mf], bf0, acc[mf][2], 0, 0, 0);
-      acc[mf][3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-          af[mf], bf1, acc[mf][3], 0, 0, 0);
-    }
-    __builtin_amdgcn_s_setprio(0);
-    // slot boundary: slot s+1 must have landed (counted wait, never 0)
-    asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+  // ---- main loop: 4-slot groups (compile-time ring ids, no tail clamp)
+  const int ns_main = nslot > 3 ? (nslot - 3) & ~3 : 0;
+  int s = 0;
+  for (; s < ns_main; s += 4) {
+    do_slot(s, 0, true);
+    do_slot(s + 1, 1, true);
+    do_slot(s + 2, 2, true);
+    do_slot(s + 3, 3, true);
   }
+  for (; s < nslot; ++s)   // tail: staging clamped to the last slot
+    do_slot(s, s & 3, s + 3 < nslot);
 
   // ---- epilogue
   const long crow0 = (long)tm * 256 + wm * 128;

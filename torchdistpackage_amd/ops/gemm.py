@@ -69,8 +69,10 @@ class _TdpaLinearFn(torch.autograd.Function):
         # fprop gating ensures M%256, N%256, K%32; both backward GEMMs
         # additionally need the in-features dim 256-aligned
         if K % 256 == 0:
-            dx = e.gemm_dgrad(dy, weight, False)
-            dw = e.gemm_wgrad(dy, x2d, pick_splitk(N, K, M), False)
+            # kswz=True: the k-outer tr16 swizzle measured +26%/+60-90%
+            # on dgrad/wgrad (gpurun_out/kbench_gemm.log r2)
+            dx = e.gemm_dgrad(dy, weight, True)
+            dw = e.gemm_wgrad(dy, x2d, pick_splitk(N, K, M), True)
         else:
             dx = dy @ weight
             dw = dy.t() @ x2d
