@@ -142,18 +142,27 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   const unsigned trJ = (unsigned)(((lane >> 2) & 3) << 6);
   const unsigned lds0 = lds_addr_of(lds);
 
+  // tr16 reads are issued WITHOUT a per-pair wait: all of a phase's
+  // gathers stay in flight together and one lgkmcnt(0) + sched_barrier(0)
+  // (methodology rule 18: hipcc hoists register-only MFMA past a standalone
+  // inline-asm wait) lands before the MFMA cluster via tr16_fence().
   auto tr16pair = [&](unsigned addr) -> bf16x8_v {
     unsigned long long lo, hi;
     asm volatile(
         "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-        "ds_read_b64_tr_b16 %1, %2 offset:2048\n\t"
-        "s_waitcnt lgkmcnt(0)"
+        "ds_read_b64_tr_b16 %1, %2 offset:2048"
         : "=v"(lo), "=v"(hi)
         : "v"(addr));
     bf16x8_v r;
     ((unsigned long long*)&r)[0] = lo;
     ((unsigned long long*)&r)[1] = hi;
     return r;
+  };
+  auto tr16_fence = [&]() {
+    if (LAYA == 1 || LAYB == 1) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+    }
   };
   auto readA = [&](int ring, int mf) -> bf16x8_v {
     if (LAYA == 0)
@@ -190,6 +199,7 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) bfr[nf] = readB(ring, nf);
     asm volatile("s_barrier" ::: "memory");
+    tr16_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mf = 0; mf < 8; ++mf) {
