@@ -1,7 +1,8 @@
 """Single-shape GEMM loops for rocprofv3 PMC collection (no timing)."""
+import os
 import sys
 import torch
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from torchdistpackage_amd.ops import ext
 
 kind = sys.argv[1] if len(sys.argv) > 1 else "fprop"

@@ -75,15 +75,21 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
                                float* __restrict__ slab,
                                int M, int N, int K,
                                long lda, long ldb, long ldc,
-                               int splitk, int kper, int tiles_n) {
+                               int splitk, int kper, int tiles_n,
+                               int gsup) {
   __shared__ __attribute__((aligned(16))) ushort lds[8 * SLOT_HW];  // 128 KiB
 
   const int nwg = gridDim.x;
   const int wgid = xcd_remap(blockIdx.x, nwg);
   const int tile = wgid / splitk;
   const int slice = wgid - tile * splitk;
-  const int tm = tile / tiles_n;
-  const int tn = tile - tm * tiles_n;
+  // L2 supertile: an XCD's contiguous wgid chunk visits tiles in
+  // gsup-tall tm bands so BOTH operand panels get L2 reuse (a row-major
+  // chunk reuses only the A panel)
+  const int band = tile / (gsup * tiles_n);
+  const int rem = tile - band * (gsup * tiles_n);
+  const int tm = band * gsup + (rem - (rem / gsup) * gsup);
+  const int tn = rem / gsup;
 
   const int tid = threadIdx.x;
   const int w = tid >> 6;            // wave id (wave-uniform)
@@ -310,6 +316,7 @@ torch::Tensor launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
   auto C = torch::empty({M, N}, A.options());
   const int tiles_m = M / 256, tiles_n = N / 256;
   const int nwg = tiles_m * tiles_n * splitk;
+  const int gsup = (tiles_m % 4 == 0) ? 4 : ((tiles_m % 2 == 0) ? 2 : 1);
   const ushort* bptr = nullptr;
   if (bias.has_value()) {
     TORCH_CHECK(bias->numel() == N && bias->scalar_type() == torch::kBFloat16);
@@ -322,7 +329,7 @@ torch::Tensor launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
     hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 0, stream,
                        (const ushort*)A.data_ptr(), (const ushort*)B.data_ptr(),
                        bptr, (ushort*)C.data_ptr(), nullptr, M, N, K,
-                       lda, ldb, (long)N, 1, K, tiles_n);
+                       lda, ldb, (long)N, 1, K, tiles_n, gsup);
   } else {
     auto slab = torch::empty({splitk, (long)M * N},
                              A.options().dtype(torch::kFloat));
@@ -331,7 +338,7 @@ torch::Tensor launch_gemm(const torch::Tensor& A, const torch::Tensor& B,
     hipLaunchKernelGGL(kern, dim3(nwg), dim3(512), 0, stream,
                        (const ushort*)A.data_ptr(), (const ushort*)B.data_ptr(),
                        nullptr, nullptr, slab.data_ptr<float>(), M, N, K,
-                       lda, ldb, (long)N, splitk, K / splitk, tiles_n);
+                       lda, ldb, (long)N, splitk, K / splitk, tiles_n, gsup);
     const long total = (long)M * N;
     const long nb = (total / 4 + 255) / 256;
     hipLaunchKernelGGL(splitk_reduce_kernel, dim3(nb), dim3(256), 0, stream,
