@@ -148,44 +148,76 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   const unsigned trJ = (unsigned)(((lane >> 2) & 3) << 6);
   const unsigned lds0 = lds_addr_of(lds);
 
-  // tr16 reads are issued WITHOUT a per-pair wait: all of a phase's
-  // gathers stay in flight together and one lgkmcnt(0) + sched_barrier(0)
-  // (methodology rule 18: hipcc hoists register-only MFMA past a standalone
-  // inline-asm wait) lands before the MFMA cluster via tr16_fence().
-  auto tr16pair = [&](unsigned addr) -> bf16x8_v {
-    unsigned long long lo, hi;
-    asm volatile(
-        "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-        "ds_read_b64_tr_b16 %1, %2 offset:2048"
-        : "=v"(lo), "=v"(hi)
-        : "v"(addr));
-    bf16x8_v r;
-    ((unsigned long long*)&r)[0] = lo;
-    ((unsigned long long*)&r)[1] = hi;
-    return r;
-  };
-  auto tr16_fence = [&]() {
-    if (LAYA == 1 || LAYB == 1) {
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
+  // ALL of an operand's tr16 gathers issue inside ONE asm block whose
+  // final instruction is the lgkm wait: the reads stay in flight together
+  // (a per-pair wait serialized 12 LDS round trips per slot), and no
+  // compiler-inserted register move can touch an output while its read is
+  // still outstanding (free-floating no-wait asm reads raced exactly that
+  // way — schedule-dependent corruption on the LAY=1 paths).
+  auto tr16x8 = [&](const unsigned* ad, bf16x8_v* fr, int n) {
+    unsigned long long r[16];
+    if (n == 8) {
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %16 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %16 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %2, %17 offset:0\n\t"
+          "ds_read_b64_tr_b16 %3, %17 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %4, %18 offset:0\n\t"
+          "ds_read_b64_tr_b16 %5, %18 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %6, %19 offset:0\n\t"
+          "ds_read_b64_tr_b16 %7, %19 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %8, %20 offset:0\n\t"
+          "ds_read_b64_tr_b16 %9, %20 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %10, %21 offset:0\n\t"
+          "ds_read_b64_tr_b16 %11, %21 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %12, %22 offset:0\n\t"
+          "ds_read_b64_tr_b16 %13, %22 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %14, %23 offset:0\n\t"
+          "ds_read_b64_tr_b16 %15, %23 offset:2048\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(r[0]), "=&v"(r[1]), "=&v"(r[2]), "=&v"(r[3]),
+            "=&v"(r[4]), "=&v"(r[5]), "=&v"(r[6]), "=&v"(r[7]),
+            "=&v"(r[8]), "=&v"(r[9]), "=&v"(r[10]), "=&v"(r[11]),
+            "=&v"(r[12]), "=&v"(r[13]), "=&v"(r[14]), "=&v"(r[15])
+          : "v"(ad[0]), "v"(ad[1]), "v"(ad[2]), "v"(ad[3]),
+            "v"(ad[4]), "v"(ad[5]), "v"(ad[6]), "v"(ad[7]));
+    } else {
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %8 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %2, %9 offset:0\n\t"
+          "ds_read_b64_tr_b16 %3, %9 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %4, %10 offset:0\n\t"
+          "ds_read_b64_tr_b16 %5, %10 offset:2048\n\t"
+          "ds_read_b64_tr_b16 %6, %11 offset:0\n\t"
+          "ds_read_b64_tr_b16 %7, %11 offset:2048\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(r[0]), "=&v"(r[1]), "=&v"(r[2]), "=&v"(r[3]),
+            "=&v"(r[4]), "=&v"(r[5]), "=&v"(r[6]), "=&v"(r[7])
+          : "v"(ad[0]), "v"(ad[1]), "v"(ad[2]), "v"(ad[3]));
+    }
+#pragma unroll
+    for (int i = 0; i < n; ++i) {
+      ((unsigned long long*)&fr[i])[0] = r[2 * i];
+      ((unsigned long long*)&fr[i])[1] = r[2 * i + 1];
     }
   };
-  auto readA = [&](int ring, int mf) -> bf16x8_v {
-    if (LAYA == 0)
-      return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdA + mf * 512];
+  auto trA_addr = [&](int ring, int mf) -> unsigned {
     const unsigned r16 = (unsigned)((wm * 128 + mf * 16) * 2);
-    const unsigned a = lds0 + (unsigned)(ring * SLOT_HW * 2) + trlane +
-                       (KSWZ ? (r16 ^ trJ) : r16);
-    return tr16pair(a);
+    return lds0 + (unsigned)(ring * SLOT_HW * 2) + trlane +
+           (KSWZ ? (r16 ^ trJ) : r16);
   };
-  auto readB = [&](int ring, int nf) -> bf16x8_v {
-    if (LAYB == 0)
-      return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdB + nf * 512];
+  auto trB_addr = [&](int ring, int nf) -> unsigned {
     const unsigned r16 = (unsigned)((wn * 64 + nf * 16) * 2);
-    const unsigned a = lds0 + (unsigned)(BREG_HW * 2) +
-                       (unsigned)(ring * SLOT_HW * 2) + trlane +
-                       (KSWZ ? (r16 ^ trJ) : r16);
-    return tr16pair(a);
+    return lds0 + (unsigned)(BREG_HW * 2) +
+           (unsigned)(ring * SLOT_HW * 2) + trlane +
+           (KSWZ ? (r16 ^ trJ) : r16);
+  };
+  auto readA0 = [&](int ring, int mf) -> bf16x8_v {
+    return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdA + mf * 512];
+  };
+  auto readB0 = [&](int ring, int nf) -> bf16x8_v {
+    return *(const bf16x8_v*)&lds[ring * SLOT_HW + rdB + nf * 512];
   };
 
   f32x4 acc[8][4];
@@ -200,12 +232,25 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   auto do_slot = [&](int s, int ring, bool stage_ok) {
     stage(Abase, lda, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1, 0, LAYA);
     bf16x8_v af[8], bfr[4];
+    if (LAYA == 0) {
 #pragma unroll
-    for (int mf = 0; mf < 8; ++mf) af[mf] = readA(ring, mf);
+      for (int mf = 0; mf < 8; ++mf) af[mf] = readA0(ring, mf);
+    } else {
+      unsigned ad[8];
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) bfr[nf] = readB(ring, nf);
+      for (int mf = 0; mf < 8; ++mf) ad[mf] = trA_addr(ring, mf);
+      tr16x8(ad, af, 8);
+    }
+    if (LAYB == 0) {
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) bfr[nf] = readB0(ring, nf);
+    } else {
+      unsigned ad[4];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) ad[nf] = trB_addr(ring, nf);
+      tr16x8(ad, bfr, 4);
+    }
     asm volatile("s_barrier" ::: "memory");
-    tr16_fence();
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mf = 0; mf < 8; ++mf) {
