@@ -250,6 +250,8 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
       for (int nf = 0; nf < 4; ++nf) ad[nf] = trB_addr(ring, nf);
       tr16x8(ad, bfr, 4);
     }
+    stage(Bbase, ldb, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1,
+          BREG_HW, LAYB);
     asm volatile("s_barrier" ::: "memory");
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -259,12 +261,6 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
       acc[mf][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
           af[mf], bfr[1], acc[mf][1], 0, 0, 0);
     }
-    __builtin_amdgcn_s_setprio(0);
-    asm volatile("s_barrier" ::: "memory");
-    stage(Bbase, ldb, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1,
-          BREG_HW, LAYB);
-    asm volatile("s_barrier" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mf = 0; mf < 8; ++mf) {
       acc[mf][2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
