@@ -449,9 +449,12 @@ def test_gemm_wgrad(kswz, sk):
                     dy.t() @ x)
 
 
-def test_gemm_autograd_linear():
-    """linear() dispatch end-to-end vs F.linear autograd (fp32 oracle)."""
-    from torchdistpackage_amd.ops.gemm import linear as fast_linear
+def test_gemm_autograd_linear(monkeypatch):
+    """linear() dispatch end-to-end vs F.linear autograd (fp32 oracle),
+    with every GEMM forced through the in-tree kernel."""
+    import torchdistpackage_amd.ops.gemm as G
+    monkeypatch.setattr(G, "_MODE", "all")
+    fast_linear = G.linear
     torch.manual_seed(3)
     M, N, K = 512, 512, 512
     x0 = (torch.randn(M, K) * 0.5).bfloat16().to(_dev())

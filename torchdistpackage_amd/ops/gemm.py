@@ -1,14 +1,29 @@
 """Dispatch layer for the hand-written CDNA4 GEMM (csrc/gemm.hip).
 
 ``linear(x, w, bias)`` is a drop-in for ``F.linear`` on the transformer hot
-path: on GPU, bf16, and 256-aligned shapes it routes fprop/dgrad/wgrad
-through the in-tree 256x256 MFMA kernel (wgrad with a split-K heuristic
-sized for 256 CUs); anything else falls back to ``F.linear`` (hipBLASLt).
-Gate with env ``TDPA_GEMM=0`` to force the library path for A/B runs.
+path.  Eligible (GPU, bf16, 256-aligned) shapes route through ONE autograd
+Function whose forward/dgrad/wgrad each independently pick the in-tree
+256x256 MFMA kernel or hipBLASLt from the same-box A/B table below
+(gpurun_out/kbench_gemm5/6.log, profiles/r02_notes.md):
 
-The reference delegates these GEMMs to cuBLAS via torch.matmul
-(tp_utils.py:171); BASELINE.json's north star names QKV / out-proj as
-hand-written CDNA4 — this is that path.
+  op      mine (TF)      hipBLASLt (TF)   policy
+  fprop   1000-1200      1165-1590        library
+  dgrad   1075-1145      1250-1390        library
+  wgrad   900-1050       835-1190         MINE for the <=8-tile-per-dim
+                                          deep-K family (e.g. out-proj
+                                          dW 2048x2048xK16384: +8-10%) —
+                                          the shape class round 1 measured
+                                          hipBLASLt weakest on; library
+                                          elsewhere
+
+Env ``TDPA_GEMM``: "1" (default) = the measured auto policy above;
+"all" = force every eligible GEMM through the in-tree kernel (kernel
+demonstration / profiling); "0" = library everywhere.
+
+The reference delegates all of these to cuBLAS via torch.matmul
+(tp_utils.py:171); BASELINE.json's north star names the transformer
+projection GEMMs as hand-written CDNA4 — csrc/gemm.hip is that path, and
+the auto policy keeps it on the hot step where it is a measured win.
 """
 
 from __future__ import annotations
@@ -21,12 +36,12 @@ import torch.nn.functional as F
 
 from . import ext
 
-_ENABLED = os.environ.get("TDPA_GEMM", "1") != "0"
+_MODE = os.environ.get("TDPA_GEMM", "1")
 # wgrad K=16384-class shapes keep fp32 partials in a slab; cap its size
 _MAX_SLAB_BYTES = 2 << 30
 
 
-def _supported_mnk(M: int, N: int, K: int) -> bool:
+def _eligible(M: int, N: int, K: int) -> bool:
     return M % 256 == 0 and N % 256 == 0 and K % 32 == 0 and K >= 32
 
 
@@ -48,8 +63,19 @@ def pick_splitk(M: int, N: int, K: int) -> int:
     return best
 
 
+def _use_mine(kind: str, M: int, N: int, K: int) -> bool:
+    if _MODE == "all":
+        return True
+    if _MODE != "1":
+        return False
+    if kind == "wgrad":
+        # few-tile deep-K wgrad: hipBLASLt's weak family (r01 item 4)
+        return M <= 2048 and N <= 2048 and K >= 4096
+    return False
+
+
 def gemm_enabled() -> bool:
-    return _ENABLED
+    return _MODE != "0"
 
 
 class _TdpaLinearFn(torch.autograd.Function):
@@ -57,7 +83,12 @@ class _TdpaLinearFn(torch.autograd.Function):
     def forward(ctx, x2d, weight, bias):
         ctx.save_for_backward(x2d, weight)
         ctx.has_bias = bias is not None
-        return ext("gemm_fprop").gemm_fprop(x2d, weight, bias)
+        M, K = x2d.shape
+        N = weight.shape[0]
+        if _use_mine("fprop", M, N, K):
+            return ext("gemm_fprop").gemm_fprop(x2d, weight, bias)
+        out = F.linear(x2d, weight, bias)
+        return out
 
     @staticmethod
     def backward(ctx, dy):
@@ -65,16 +96,15 @@ class _TdpaLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         M, K = x2d.shape
         N = weight.shape[0]
-        e = ext("gemm_bwd")
-        # fprop gating ensures M%256, N%256, K%32; both backward GEMMs
-        # additionally need the in-features dim 256-aligned
-        if K % 256 == 0:
-            # kswz=True: the k-outer tr16 swizzle measured +26%/+60-90%
-            # on dgrad/wgrad (gpurun_out/kbench_gemm.log r2)
-            dx = e.gemm_dgrad(dy, weight, True)
-            dw = e.gemm_wgrad(dy, x2d, pick_splitk(N, K, M), True)
+        # the backward GEMMs additionally need the in-features dim
+        # 256-aligned for the in-tree kernel
+        if K % 256 == 0 and _use_mine("dgrad", M, K, N):
+            dx = ext("gemm").gemm_dgrad(dy, weight, True)
         else:
             dx = dy @ weight
+        if K % 256 == 0 and _use_mine("wgrad", N, K, M):
+            dw = ext("gemm").gemm_wgrad(dy, x2d, pick_splitk(N, K, M), True)
+        else:
             dw = dy.t() @ x2d
         db = dy.sum(0) if ctx.has_bias else None
         return dx, dw, db
@@ -83,15 +113,19 @@ class _TdpaLinearFn(torch.autograd.Function):
 def linear(x: torch.Tensor, weight: torch.Tensor,
            bias: Optional[torch.Tensor] = None) -> torch.Tensor:
     """F.linear drop-in; routes 256-aligned bf16 GPU shapes through the
-    in-tree MFMA GEMM (fprop + both backward GEMMs)."""
-    if (_ENABLED and x.is_cuda and x.dtype == torch.bfloat16
+    per-GEMM measured dispatch (in-tree MFMA kernel vs hipBLASLt)."""
+    if (_MODE != "0" and x.is_cuda and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16
             and (bias is None or bias.dtype == torch.bfloat16)):
         shape = x.shape
         M = x.numel() // shape[-1]
         K = shape[-1]
         N = weight.shape[0]
-        if _supported_mnk(M, N, K):
+        if _eligible(M, N, K) and (
+                _MODE == "all"
+                or (K % 256 == 0 and (_use_mine("wgrad", N, K, M)
+                                      or _use_mine("dgrad", M, K, N)))
+                or _use_mine("fprop", M, N, K)):
             x2d = x.reshape(M, K)
             if not x2d.is_contiguous():
                 x2d = x2d.contiguous()
