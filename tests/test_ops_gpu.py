@@ -477,3 +477,49 @@ def test_gemm_autograd_linear(monkeypatch):
                    (b1.grad, b2.grad)]:
         e = (g1.float() - g2).abs().max() / g2.abs().max().clamp_min(1e-6)
         assert e.item() < 0.06, e.item()
+
+
+# ------------------------------------------------------- RoPE and SwiGLU
+
+def test_rope_kernel():
+    from torchdistpackage_amd.ops import rope_rotate_half
+    torch.manual_seed(4)
+    B, H, S, D = 2, 4, 64, 128
+    inv = 1.0 / (10000.0 ** (torch.arange(0, D, 2).float() / D))
+    freqs = torch.outer(torch.arange(S + 8).float(), inv)
+    cos, sin = freqs.cos().to(_dev()), freqs.sin().to(_dev())
+    for pos0 in (0, 8):
+        x = (torch.randn(B, H, S, D) * 0.5).bfloat16().to(_dev())
+        x1 = x.clone().requires_grad_(True)
+        y = rope_rotate_half(x1, cos, sin, pos0)
+        # fp32 oracle
+        xf = x.float().requires_grad_(True)
+        c = cos[pos0:pos0 + S]
+        sn = sin[pos0:pos0 + S]
+        a, b = xf[..., :D // 2], xf[..., D // 2:]
+        ref = torch.cat([a * c - b * sn, b * c + a * sn], dim=-1)
+        assert (y.float() - ref).abs().max() < 0.02
+        g = torch.randn_like(ref)
+        y.backward(g.bfloat16())
+        ref.backward(g)
+        assert (x1.grad.float() - xf.grad).abs().max() < 0.02
+
+
+def test_swiglu_kernel():
+    from torchdistpackage_amd.ops import swiglu
+    torch.manual_seed(5)
+    a0 = (torch.randn(512, 256) * 2).bfloat16().to(_dev())
+    b0 = (torch.randn(512, 256) * 2).bfloat16().to(_dev())
+    a1, b1 = a0.clone().requires_grad_(True), b0.clone().requires_grad_(True)
+    out = swiglu(a1, b1)
+    af = a0.float().requires_grad_(True)
+    bf = b0.float().requires_grad_(True)
+    ref = torch.nn.functional.silu(af) * bf
+    assert (out.float() - ref).abs().max() / ref.abs().max() < 0.02
+    g = torch.randn_like(ref)
+    out.backward(g.bfloat16())
+    ref.backward(g)
+    assert (a1.grad.float() - af.grad).abs().max() / \
+        af.grad.abs().max().clamp_min(1e-6) < 0.03
+    assert (b1.grad.float() - bf.grad).abs().max() / \
+        bf.grad.abs().max().clamp_min(1e-6) < 0.03

@@ -20,7 +20,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import RMSNorm, flash_attention
+from ..ops import RMSNorm, flash_attention, rope_rotate_half, swiglu
 from ..ops.gemm import linear as fast_linear
 from ..parallel.tensor.tp_utils import (ColParallelLinear, RowParallelLinear,
                                         copy_to_tp_region,
@@ -72,12 +72,8 @@ class Rope(nn.Module):
         self.register_buffer("sin", freqs.sin(), persistent=False)
 
     def forward(self, x: torch.Tensor, pos0: int = 0) -> torch.Tensor:
-        # x (B, H, S, hd)
-        S, hd = x.shape[-2], x.shape[-1]
-        cos = self.cos[pos0:pos0 + S].to(x.dtype)       # (S, hd/2)
-        sin = self.sin[pos0:pos0 + S].to(x.dtype)
-        x1, x2 = x[..., :hd // 2], x[..., hd // 2:]
-        return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+        # x (B, H, S, hd) — fused rotate-half kernel (ops/csrc/rope_swiglu.hip)
+        return rope_rotate_half(x, self.cos, self.sin, pos0)
 
 
 class LlamaAttention(nn.Module):
@@ -144,8 +140,8 @@ class LlamaMlp(nn.Module):
             x = gather_from_sequence_parallel_region(x)
         else:
             x = copy_to_tp_region(x)
-        return self.w2(F.silu(fast_linear(x, self.w1.weight)) *
-                       fast_linear(x, self.w3.weight))
+        return self.w2(swiglu(fast_linear(x, self.w1.weight),
+                              fast_linear(x, self.w3.weight)))
 
 
 class LlamaBlock(nn.Module):

@@ -475,3 +475,77 @@ def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor):
         return _CrossEntropyFn.apply(l2.contiguous(), t)
     import torch.nn.functional as F
     return F.cross_entropy(l2.float(), t)
+
+
+# ---------------------------------------------------------------------------
+# RoPE (rotate-half) and SwiGLU — fused Llama hot-path ops
+# ---------------------------------------------------------------------------
+
+class _RopeFn(torch.autograd.Function):
+    """Rotate-half RoPE over (B, H, S, D) with fp32 (S, D/2) tables."""
+
+    @staticmethod
+    def forward(ctx, x, cos, sin, pos0):
+        ctx.save_for_backward(cos, sin)
+        ctx.pos0 = pos0
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            return ext("rope").rope_apply(x.contiguous(), cos, sin, pos0,
+                                          True)
+        hd = x.shape[-1]
+        S = x.shape[-2]
+        c = cos[pos0:pos0 + S].to(x.dtype)
+        s = sin[pos0:pos0 + S].to(x.dtype)
+        x1, x2 = x[..., :hd // 2], x[..., hd // 2:]
+        return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        if dy.is_cuda and dy.dtype == torch.bfloat16:
+            return (ext("rope").rope_apply(dy.contiguous(), cos, sin,
+                                           ctx.pos0, False),
+                    None, None, None)
+        hd = dy.shape[-1]
+        S = dy.shape[-2]
+        c = cos[ctx.pos0:ctx.pos0 + S].to(dy.dtype)
+        s = sin[ctx.pos0:ctx.pos0 + S].to(dy.dtype)
+        d1, d2 = dy[..., :hd // 2], dy[..., hd // 2:]
+        return (torch.cat([d1 * c + d2 * s, d2 * c - d1 * s], dim=-1),
+                None, None, None)
+
+
+def rope_rotate_half(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                     pos0: int = 0) -> torch.Tensor:
+    """Fused rotate-half rotary embedding (one read+write pass; the eager
+    form is 2 cats + 4 muls per call, models/llama.py r01)."""
+    return _RopeFn.apply(x, cos, sin, pos0)
+
+
+class _SwiGluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        ctx.save_for_backward(a, b)
+        if a.is_cuda and a.dtype == torch.bfloat16:
+            return ext("swiglu").swiglu_fwd(a.contiguous(), b.contiguous())
+        af = a.float()
+        return (af * torch.sigmoid(af) * b.float()).to(a.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        if a.is_cuda and a.dtype == torch.bfloat16:
+            da, db = ext("swiglu").swiglu_bwd(dy.contiguous(),
+                                              a.contiguous(), b.contiguous())
+            return da, db
+        af, bf, dyf = a.float(), b.float(), dy.float()
+        sig = torch.sigmoid(af)
+        silu = af * sig
+        da = dyf * bf * (sig + silu * (1 - sig))
+        db = dyf * silu
+        return da.to(a.dtype), db.to(b.dtype)
+
+
+def swiglu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """silu(a) * b, fused (backward recomputes sigmoid — nothing extra
+    saved beyond a and b)."""
+    return _SwiGluFn.apply(a, b)

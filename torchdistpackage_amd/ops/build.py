@@ -25,6 +25,7 @@ SOURCES = [
     "cross_entropy.hip",
     "probe.hip",
     "gemm.hip",
+    "rope_swiglu.hip",
 ]
 
 

@@ -1116,10 +1116,23 @@ void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                      torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
                      bool causal, double scale) {
   static const bool recompute = std::getenv("TDPA_DQ_RECOMPUTE") != nullptr;
-  if (recompute) {
-    attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
-    attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
-    return;
+  // The dS-store path buys -13% backward time (r01 v12) at the cost of an
+  // O(S^2) bf16 workspace (B*H*S*S_pad).  That grows quadratically with
+  // sequence length (536 MB at the bench shape, ~8.6 GB at S=8192/B=16) —
+  // above a fixed budget switch to the recompute-dq path, which needs no
+  // workspace and loses only the 13% (bound the
+  // workspace for the long-context direction).
+  constexpr long kMaxDsBytes = 4L << 30;  // 4 GiB cap (S=4096/B=4 stays
+                                          // on the fast path, S>=8192 not)
+  {
+    const long Sq = q.size(2);
+    const long spad_est = (Sq + QT2 - 1) / QT2 * QT2;
+    const long ds_bytes = (long)q.size(0) * q.size(1) * Sq * spad_est * 2;
+    if (recompute || ds_bytes > kMaxDsBytes) {
+      attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
+      attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
+      return;
+    }
   }
   const int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   const int Hkv = k.size(1);

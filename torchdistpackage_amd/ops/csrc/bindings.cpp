@@ -41,6 +41,11 @@ torch::Tensor tr16_probe(long addr_mode);
 std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor targets);
 torch::Tensor gemm_fprop(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias);
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
+                         long pos0, bool fwd);
+torch::Tensor swiglu_fwd(torch::Tensor a, torch::Tensor b);
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dy, torch::Tensor a,
+                                      torch::Tensor b);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w, bool kswz);
 torch::Tensor gemm_wgrad(torch::Tensor dy, torch::Tensor x, long splitk,
                          bool kswz);
@@ -66,6 +71,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tr16_probe", &tr16_probe);
   m.def("ce_fwd", &ce_fwd);
   m.def("gemm_fprop", &gemm_fprop);
+  m.def("rope_apply", &rope_apply);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
   m.def("gemm_dgrad", &gemm_dgrad);
   m.def("gemm_wgrad", &gemm_wgrad);
   m.def("ce_bwd", &ce_bwd);

@@ -1,0 +1,182 @@
+// Fused RoPE (rotate-half) and SwiGLU kernels for the Llama hot path.
+//
+// Replaces the eager chains the round-1 model used
+// (models/llama.py: rotate-half = 2 cats + 4 muls per call per layer;
+// silu(w1x)*w3x unfused) — both are HBM-bound elementwise ops, so the win
+// is one read+write pass instead of 4-6 (plus the launch count).
+// BASELINE.json's north star names RoPE as a hand-written CDNA4 kernel.
+//
+// Conventions: bf16 tensors, fp32 cos/sin tables (S, D/2), fp32 math,
+// ushort8 (16 B/lane) accesses — the bandwidth rules from the guide that
+// took bias_gelu to the roofline in round 1.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+
+DEVINL void load8(const unsigned short* p, float* f) {
+  ushort4 a = *(const ushort4*)p;
+  ushort4 b = *(const ushort4*)(p + 4);
+  f[0] = bf2f(a.x); f[1] = bf2f(a.y); f[2] = bf2f(a.z); f[3] = bf2f(a.w);
+  f[4] = bf2f(b.x); f[5] = bf2f(b.y); f[6] = bf2f(b.z); f[7] = bf2f(b.w);
+}
+
+DEVINL void store8(unsigned short* p, const float* f) {
+  ushort4 a, b;
+  a.x = f2bf(f[0]); a.y = f2bf(f[1]); a.z = f2bf(f[2]); a.w = f2bf(f[3]);
+  b.x = f2bf(f[4]); b.y = f2bf(f[5]); b.z = f2bf(f[6]); b.w = f2bf(f[7]);
+  *(ushort4*)p = a;
+  *(ushort4*)(p + 4) = b;
+}
+
+// rows = B*H*S (seq position = row % S), rotate-half pairs (i, i+D/2).
+// FWD=false computes the transpose rotation (gradient).
+template <bool FWD>
+__global__ void rope_kernel(const unsigned short* __restrict__ x,
+                            unsigned short* __restrict__ y,
+                            const float* __restrict__ cs,   // (S, D/2) cos
+                            const float* __restrict__ sn,   // (S, D/2) sin
+                            long rows, int S, int D, int pos0) {
+  const int half = D >> 1;
+  const int chunks = half >> 3;                 // 8 elems per thread-chunk
+  const long total = rows * chunks;
+  for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total;
+       g += (long)gridDim.x * blockDim.x) {
+    const long row = g / chunks;
+    const int c = (int)(g - row * chunks) * 8;
+    const int s = (int)(row % S) + pos0;
+    const long base = row * D + c;
+    float x1[8], x2[8], co[8], si[8];
+    load8(x + base, x1);
+    load8(x + base + half, x2);
+#pragma unroll
+    for (int j = 0; j < 8; j += 4) {
+      *(float4*)(co + j) = *(const float4*)(cs + (long)s * half + c + j);
+      *(float4*)(si + j) = *(const float4*)(sn + (long)s * half + c + j);
+    }
+    float y1[8], y2[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (FWD) {
+        y1[j] = x1[j] * co[j] - x2[j] * si[j];
+        y2[j] = x2[j] * co[j] + x1[j] * si[j];
+      } else {   // transpose (inverse rotation): grads
+        y1[j] = x1[j] * co[j] + x2[j] * si[j];
+        y2[j] = x2[j] * co[j] - x1[j] * si[j];
+      }
+    }
+    store8(y + base, y1);
+    store8(y + base + half, y2);
+  }
+}
+
+// silu(a) * b and its backward (recomputes sigmoid from a: saves storing
+// the activation).  exp via the hardware exp2 unit (log2e-scaled).
+DEVINL float sigmoidf_fast(float x) {
+  return 1.0f / (1.0f + __builtin_exp2f(-1.442695041f * x));
+}
+
+__global__ void swiglu_fwd_kernel(const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b,
+                                  unsigned short* __restrict__ out,
+                                  long n) {
+  const long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (g >= n) return;
+  float fa[8], fb[8], fo[8];
+  load8(a + g, fa);
+  load8(b + g, fb);
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    fo[j] = fa[j] * sigmoidf_fast(fa[j]) * fb[j];
+  store8(out + g, fo);
+}
+
+__global__ void swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
+                                  const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b,
+                                  unsigned short* __restrict__ da,
+                                  unsigned short* __restrict__ db,
+                                  long n) {
+  const long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (g >= n) return;
+  float fdy[8], fa[8], fb[8], fda[8], fdb[8];
+  load8(dy + g, fdy);
+  load8(a + g, fa);
+  load8(b + g, fb);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const float sig = sigmoidf_fast(fa[j]);
+    const float silu = fa[j] * sig;
+    fda[j] = fdy[j] * fb[j] * (sig + silu * (1.0f - sig));
+    fdb[j] = fdy[j] * silu;
+  }
+  store8(da + g, fda);
+  store8(db + g, fdb);
+}
+
+void check_bf16(const torch::Tensor& t, const char* n) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
+              t.is_contiguous(), n, " must be contiguous bf16 CUDA");
+}
+
+}  // namespace
+
+// x (B, H, S, D) bf16 contiguous; cos/sin (>=S+pos0, D/2) fp32.
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
+                         long pos0, bool fwd) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 4 && cs.scalar_type() == torch::kFloat &&
+              cs.is_contiguous() && sn.is_contiguous());
+  const int D = x.size(3), S = x.size(2);
+  TORCH_CHECK(D % 16 == 0 && cs.size(1) == D / 2);
+  const long rows = (long)x.size(0) * x.size(1) * S;
+  auto y = torch::empty_like(x);
+  const long total = rows * (D / 16);
+  const long nb = (total + BLOCK - 1) / BLOCK;
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto kern = fwd ? rope_kernel<true> : rope_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3((unsigned)std::min(nb, (long)65535 * 8)),
+                     dim3(BLOCK), 0, stream,
+                     (const unsigned short*)x.data_ptr(),
+                     (unsigned short*)y.data_ptr(), cs.data_ptr<float>(),
+                     sn.data_ptr<float>(), rows, S, D, (int)pos0);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a"); check_bf16(b, "b");
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % 8 == 0);
+  auto out = torch::empty_like(a);
+  const long n = a.numel();
+  const long nb = (n / 8 + BLOCK - 1) / BLOCK;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3((unsigned)nb), dim3(BLOCK), 0,
+                     stream, (const unsigned short*)a.data_ptr(),
+                     (const unsigned short*)b.data_ptr(),
+                     (unsigned short*)out.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return out;
+}
+
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dy, torch::Tensor a,
+                                      torch::Tensor b) {
+  check_bf16(dy, "dy"); check_bf16(a, "a"); check_bf16(b, "b");
+  auto da = torch::empty_like(a);
+  auto db = torch::empty_like(b);
+  const long n = a.numel();
+  const long nb = (n / 8 + BLOCK - 1) / BLOCK;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3((unsigned)nb), dim3(BLOCK), 0,
+                     stream, (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)a.data_ptr(),
+                     (const unsigned short*)b.data_ptr(),
+                     (unsigned short*)da.data_ptr(),
+                     (unsigned short*)db.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return {da, db};
+}
