@@ -63,32 +63,7 @@ DEVINL unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
-// One 32x32-MFMA A-fragment gathered from a BLOCKED natural image
-// img[d/16][row][16] by two ds_read_b64_tr_b16 (the probe-verified model;
-// the dq-lite kernel below pioneered it, r01 +30% on that kernel).  The
-// wait stays INSIDE the asm: a free-floating no-wait read raced with
-// register moves in the GEMM (profiles/r02_notes.md).
-DEVINL bf16x8_v tr16_afrag(unsigned addr) {
-  unsigned long long lo, hic;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-      "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=v"(lo), "=v"(hic)
-      : "v"(addr));
-  bf16x8_v r;
-  ((unsigned long long*)&r)[0] = lo;
-  ((unsigned long long*)&r)[1] = hic;
-  return r;
-}
-
-// per-lane byte base of the tr16 gather within one [8][64][16] image
-DEVINL unsigned tr16_lane_base(int lane) {
-  return 2u * (((lane >> 4) & 1) * 1024u + ((lane >> 2) & 3) * 16u +
-               (lane & 3) * 4u + (lane >> 5) * 128u);
-}
-
-template <bool CAUSAL, bool TRT>
+template <bool CAUSAL>
 __launch_bounds__(512)
 __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
                                    const unsigned short* __restrict__ k,
@@ -120,9 +95,6 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
   const unsigned short* qp = q + bb * qs.b + hh * qs.h;
   const unsigned short* kp = k + bb * ks.b + hkv * ks.h;
   const unsigned short* vp = v + bb * vs.b + hkv * vs.h;
-  const unsigned trv_base = (unsigned)(unsigned long long)
-      (__attribute__((address_space(3))) unsigned short*)&vt_lds[0][0] +
-      tr16_lane_base(lane);
 
   // ---- Q fragments: this lane's q-row, 8 chunks of 16 (B-operand layout)
   const int qrow = qbase + wid * QW + l31;
@@ -170,15 +142,10 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
       int key = idx / D2;
       int col = idx % D2;
       *(bf16x8_v*)&k_lds[0][swzK(key, col)] = k_reg[c];
-      if (TRT) {
-        *(bf16x8_v*)&vt_lds[0][(col >> 4) * (KV * 16) + key * 16 +
-                               (col & 15)] = v_reg[c];
-      } else {
-        const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
+      const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vt_lds[0][swzV(col + j, key)] = vsrc[j];
-      }
+      for (int j = 0; j < 8; ++j)
+        vt_lds[0][swzV(col + j, key)] = vsrc[j];
     }
     if (KV < kv_end) stage_load(KV);
     __syncthreads();
@@ -289,14 +256,8 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
       for (int kt = 0; kt < 2; ++kt) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8_v a_v;
-          if (TRT)
-            a_v = tr16_afrag(trv_base + (unsigned)(buf * 2 * D2 * KV) +
-                             (unsigned)(ds * 4096 +
-                                        (kt * 32 + kc * 16) * 32));
-          else
-            a_v = pack8v(&vt_lds[buf][swzV(ds * 32 + l31,
-                                           kt * 32 + kc * 16 + hi * 8)]);
+          bf16x8_v a_v = pack8v(&vt_lds[buf][swzV(ds * 32 + l31,
+                                                  kt * 32 + kc * 16 + hi * 8)]);
           acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_v, pb[kt][kc], acc[ds], 0, 0, 0);
         }
@@ -312,15 +273,10 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
         int key = idx / D2;
         int col = idx % D2;
         *(bf16x8_v*)&k_lds[buf ^ 1][swzK(key, col)] = k_reg[c];
-        if (TRT) {
-          *(bf16x8_v*)&vt_lds[buf ^ 1][(col >> 4) * (KV * 16) + key * 16 +
-                                       (col & 15)] = v_reg[c];
-        } else {
-          const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
+        const unsigned short* vsrc = (const unsigned short*)&v_reg[c];
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vt_lds[buf ^ 1][swzV(col + j, key)] = vsrc[j];
-        }
+        for (int j = 0; j < 8; ++j)
+          vt_lds[buf ^ 1][swzV(col + j, key)] = vsrc[j];
       }
       if (kt0 + 2 * KV < kv_end) stage_load(kt0 + 2 * KV);
     }
@@ -364,21 +320,22 @@ void attn_fwd_v2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   };
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
-  static const bool no_tr16 = std::getenv("TDPA_NO_TR16") != nullptr;
-#define L_FWD(CC, TRT)                                                        \
-  hipLaunchKernelGGL((attn_fwd_v2_kernel<CC, TRT>), grid, block, 0, stream,   \
-                     (const unsigned short*)q.data_ptr(),                     \
-                     (const unsigned short*)k.data_ptr(),                     \
-                     (const unsigned short*)v.data_ptr(),                     \
-                     (unsigned short*)o.data_ptr(),                           \
-                     lse.data_ptr<float>(), get(q), get(k), get(v), get(o),   \
-                     B, H, S, (float)scale, q_per_kv)
-  if (causal) {
-    if (no_tr16) L_FWD(true, false); else L_FWD(true, true);
-  } else {
-    if (no_tr16) L_FWD(false, false); else L_FWD(false, true);
-  }
-#undef L_FWD
+  if (causal)
+    hipLaunchKernelGGL((attn_fwd_v2_kernel<true>), grid, block, 0, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(),
+                       lse.data_ptr<float>(), get(q), get(k), get(v), get(o),
+                       B, H, S, (float)scale, q_per_kv);
+  else
+    hipLaunchKernelGGL((attn_fwd_v2_kernel<false>), grid, block, 0, stream,
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (unsigned short*)o.data_ptr(),
+                       lse.data_ptr<float>(), get(q), get(k), get(v), get(o),
+                       B, H, S, (float)scale, q_per_kv);
   HIP_CHECK_LAST();
 }
 
@@ -621,7 +578,7 @@ namespace {
 // coalesced (each 32-lane half writes one q-row, 32 consecutive keys).  The
 // dq-lite kernel then consumes dS directly instead of recomputing
 // S/P/dP per tile (drops 32 of dq's 48 MFMAs plus its V/dO staging).
-template <bool CAUSAL, bool DK_PASS, bool WRITE_DS, bool TRT>
+template <bool CAUSAL, bool DK_PASS, bool WRITE_DS>
 __launch_bounds__(512)
 __global__ void attn_bwd_dkdv_v2_kernel(
     const unsigned short* __restrict__ q,
@@ -661,9 +618,6 @@ __global__ void attn_bwd_dkdv_v2_kernel(
   const unsigned short* dop = dout + bb * dos.b + hh * dos.h;
   const float* lsep = lse + (long)bh * S;
   const float* delp = delta + (long)bh * S;
-  const unsigned tr_base = (unsigned)(unsigned long long)
-      (__attribute__((address_space(3))) unsigned short*)&tr_lds[0][0] +
-      tr16_lane_base(lane);
 
   // this wave's 32 keys: K^T fragments always; V^T only in the dk pass
   const int key = kbase + wid * QW + l31;
@@ -732,20 +686,12 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       *(bf16x8_v*)&q_lds[b][swzK(row, col)] = qv;
       if (DK_PASS)
         *(bf16x8_v*)&do_lds[b * KV * D2 + swzK(row, col)] = dv8;
-      // transposed tile: dO^T for the dv pass, Q^T for the dk pass.
-      // TRT: ONE b128 write into the blocked natural image
-      // [d/16][row][16] read back by tr16 gathers (replaces the 8-write
-      // scalar scatter per chunk)
-      if (TRT) {
-        *(bf16x8_v*)&tr_lds[b][(col >> 4) * (KV * 16) + row * 16 +
-                               (col & 15)] = DK_PASS ? qv : dv8;
-      } else {
-        const unsigned short* tsrc = DK_PASS
-            ? (const unsigned short*)&qv : (const unsigned short*)&dv8;
+      // transposed tile: dO^T for the dv pass, Q^T for the dk pass
+      const unsigned short* tsrc = DK_PASS
+          ? (const unsigned short*)&qv : (const unsigned short*)&dv8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          tr_lds[b][swzV(col + j, row)] = tsrc[j];
-      }
+      for (int j = 0; j < 8; ++j)
+        tr_lds[b][swzV(col + j, row)] = tsrc[j];
     }
     if (tid < KV) {
       int grow = qt0 + tid;
@@ -819,15 +765,9 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       for (int ds = 0; ds < 4; ++ds) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
-          bf16x8_v a_tr;
-          if (TRT)
-            a_tr = tr16_afrag(tr_base + (unsigned)(buf * 2 * D2 * KV) +
-                              (unsigned)(ds * 4096 +
-                                         (qt * 32 + kc * 16) * 32));
-          else
-            a_tr = pack8v(&tr_lds[buf][swzV(ds * 32 + l31,
-                                            qt * 32 + kc * 16 +
-                                            hi * 8)]);
+          bf16x8_v a_tr = pack8v(&tr_lds[buf][swzV(ds * 32 + l31,
+                                                   qt * 32 + kc * 16 +
+                                                   hi * 8)]);
           acc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               a_tr, fb[kc], acc[ds], 0, 0, 0);
         }
@@ -875,10 +815,9 @@ void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   };
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
-#define L_KV2(CC, WDS, DSP, SPAD, TRT)                                        \
+#define L_KV2(CC, WDS, DSP, SPAD)                                             \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false, false, TRT>),      \
-                       grid,                                                  \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false, false>), grid,     \
                        block, 0,                                              \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
@@ -888,8 +827,8 @@ void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (unsigned short*)dv.data_ptr(), nullptr, 0, get(q),    \
                        get(k), get(v), get(dout), get(dv), B, H, S,           \
                        (float)scale, q_per_kv);                               \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true, WDS, TRT>),         \
-                       grid, block, 0,                                        \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true, WDS>), grid,        \
+                       block, 0,                                              \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
                        (const unsigned short*)v.data_ptr(),                   \
@@ -899,14 +838,8 @@ void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        get(k), get(v), get(dout), get(dk), B, H, S,           \
                        (float)scale, q_per_kv);                               \
   } while (0)
-  static const bool no_tr16_kv = std::getenv("TDPA_NO_TR16") != nullptr;
-  if (causal) {
-    if (no_tr16_kv) L_KV2(true, false, nullptr, 0, false);
-    else L_KV2(true, false, nullptr, 0, true);
-  } else {
-    if (no_tr16_kv) L_KV2(false, false, nullptr, 0, false);
-    else L_KV2(false, false, nullptr, 0, true);
-  }
+  if (causal) L_KV2(true, false, nullptr, 0);
+  else L_KV2(false, false, nullptr, 0);
   HIP_CHECK_LAST();
 }
 
@@ -1213,15 +1146,10 @@ void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   const long spad = (long)grid.x * QT2;
   auto dsw = torch::empty({(long)B * H, (long)S, spad}, q.options());
   unsigned short* dsp = (unsigned short*)dsw.data_ptr();
-  static const bool no_tr16 = std::getenv("TDPA_NO_TR16") != nullptr;
-  if (causal) {
-    if (no_tr16) L_KV2(true, true, dsp, spad, false);
-    else L_KV2(true, true, dsp, spad, true);
-  } else {
-    if (no_tr16) L_KV2(false, true, dsp, spad, false);
-    else L_KV2(false, true, dsp, spad, true);
-  }
+  if (causal) L_KV2(true, true, dsp, spad);
+  else L_KV2(false, true, dsp, spad);
 #undef L_KV2
+  static const bool no_tr16 = std::getenv("TDPA_NO_TR16") != nullptr;
 #define L_DQL(CC)                                                             \
   do {                                                                        \
     if (no_tr16)                                                              \
