@@ -523,3 +523,23 @@ def test_swiglu_kernel():
         af.grad.abs().max().clamp_min(1e-6) < 0.03
     assert (b1.grad.float() - bf.grad).abs().max() / \
         bf.grad.abs().max().clamp_min(1e-6) < 0.03
+
+
+def test_flash_attention_headdim_fallback():
+    """head_dim without a HIP kernel (96) must run (composite path) and
+    match the fp32 oracle — round 1 TORCH_CHECK-failed these dims."""
+    from torchdistpackage_amd.ops import flash_attention
+    torch.manual_seed(7)
+    B, H, S, D = 2, 3, 128, 96
+    q = (torch.randn(B, H, S, D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    k = (torch.randn(B, H, S, D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    v = (torch.randn(B, H, S, D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    o = flash_attention(q, k, v, causal=True)
+    qf, kf, vf = (t.detach().float().requires_grad_(True) for t in (q, k, v))
+    s = qf @ kf.transpose(-1, -2) / (D ** 0.5)
+    mask = torch.ones(S, S, dtype=torch.bool, device=_dev()).tril_()
+    ref = torch.softmax(s.masked_fill(~mask, float("-inf")), -1) @ vf
+    assert (o.float() - ref).abs().max() < 3e-2
+    o.sum().backward()
+    ref.sum().backward()
+    assert (q.grad.float() - qf.grad).abs().max() < 0.1

@@ -305,7 +305,28 @@ class _FlashAttentionFn(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
+_warned_hd = set()
+
+
 def flash_attention(q, k, v, causal: bool = True, scale: float = None):
+    """Flash attention; the in-tree HIP kernels cover head_dim 64 and 128.
+    Other head dims (80/96/...) go through an EXPLICIT composite path
+    (torch SDPA with fp32 math) — correct but unfused; round-1 hard-failed
+    these.  The fallback is announced once so a GPU
+    run can never silently lose the kernels for supported dims."""
+    D = q.shape[-1]
+    if q.is_cuda and D not in (64, 128):
+        if D not in _warned_hd:
+            _warned_hd.add(D)
+            print(f"[tdpa.ops] flash_attention: head_dim {D} has no HIP "
+                  f"kernel (64/128 do); using composite SDPA for it")
+        rep = q.shape[1] // k.shape[1]
+        kk = k if rep == 1 else k.repeat_interleave(rep, 1)
+        vv = v if rep == 1 else v.repeat_interleave(rep, 1)
+        if scale is None:
+            scale = 1.0 / math.sqrt(D)
+        return torch.nn.functional.scaled_dot_product_attention(
+            q, kk, vv, is_causal=causal, scale=scale)
     return _FlashAttentionFn.apply(q, k, v, causal, scale)
 
 
