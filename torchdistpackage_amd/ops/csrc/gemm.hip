@@ -229,7 +229,10 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   // one 32-deep slot: 2 phases.  All 12 fragment reads issue at the top
   // (counted lgkm waits let phase b's operands arrive under phase a's
   // MFMAs); staging for slot s+3 is split across the two phases.
-  auto do_slot = [&](int s, int ring, bool stage_ok) {
+  // wait_mode: 0 = no vmcnt at this slot boundary (covered by the
+  // previous even-slot wait), 1 = vmcnt(4) (even slots: guarantees the
+  // NEXT TWO slots' staging landed), 2 = vmcnt(8) every slot (tail-safe)
+  auto do_slot = [&](int s, int ring, bool stage_ok, int wait_mode) {
     stage(Abase, lda, (ring + 3) & 3, stage_ok ? s + 3 : nslot - 1, 0, LAYA);
     bf16x8_v af[8], bfr[4];
     if (LAYA == 0) {
@@ -269,8 +272,13 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
           af[mf], bfr[3], acc[mf][3], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-    // slot boundary: slot s+1 must have landed (counted wait, never 0)
-    asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+    // slot boundary (counted waits, never 0 in the loop)
+    if (wait_mode == 1)
+      asm volatile("s_waitcnt vmcnt(4)\n\ts_barrier" ::: "memory");
+    else if (wait_mode == 2)
+      asm volatile("s_waitcnt vmcnt(8)\n\ts_barrier" ::: "memory");
+    else
+      asm volatile("s_barrier" ::: "memory");
   };
 
   // ---- prologue: slots 0..2 staged; wait until slot 0 landed
@@ -283,13 +291,13 @@ __global__ void gemm256_kernel(const ushort* __restrict__ A,
   const int ns_main = nslot > 3 ? (nslot - 3) & ~3 : 0;
   int s = 0;
   for (; s < ns_main; s += 4) {
-    do_slot(s, 0, true);
-    do_slot(s + 1, 1, true);
-    do_slot(s + 2, 2, true);
-    do_slot(s + 3, 3, true);
+    do_slot(s, 0, true, 1);
+    do_slot(s + 1, 1, true, 0);
+    do_slot(s + 2, 2, true, 1);
+    do_slot(s + 3, 3, true, 1);
   }
   for (; s < nslot; ++s)   // tail: staging clamped to the last slot
-    do_slot(s, s & 3, s + 3 < nslot);
+    do_slot(s, s & 3, s + 3 < nslot, 2);
 
   // ---- epilogue
   const long crow0 = (long)tm * 256 + wm * 128;
