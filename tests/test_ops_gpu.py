@@ -543,3 +543,24 @@ def test_flash_attention_headdim_fallback():
     o.sum().backward()
     ref.sum().backward()
     assert (q.grad.float() - qf.grad).abs().max() < 0.1
+
+
+def test_ce_partial_fwd_kernel():
+    """Fused vocab-parallel CE primitives vs the eager shard math (tp=1
+    degenerate: full range; plus an offset shard with out-of-range
+    targets)."""
+    e = ops.ext("ce_partial")
+    torch.manual_seed(8)
+    N, Vp = 64, 512
+    logits = (torch.randn(N, Vp) * 2).bfloat16().to(_dev())
+    # half the targets out of shard (-1)
+    t = torch.randint(0, Vp, (N,), device=_dev())
+    t[::2] = -1
+    lse, tgt = e.ce_partial_fwd(logits, t)
+    x = logits.float()
+    ref_lse = torch.logsumexp(x, -1)
+    assert (lse - ref_lse).abs().max() < 1e-3
+    ref_tgt = torch.where(t >= 0, x.gather(-1, t.clamp_min(0)
+                                           .unsqueeze(-1)).squeeze(-1),
+                          torch.zeros_like(ref_lse))
+    assert (tgt - ref_tgt).abs().max() < 1e-3

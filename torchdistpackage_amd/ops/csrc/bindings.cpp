@@ -51,6 +51,8 @@ torch::Tensor gemm_wgrad(torch::Tensor dy, torch::Tensor x, long splitk,
                          bool kswz);
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);
+std::vector<torch::Tensor> ce_partial_fwd(torch::Tensor logits,
+                                          torch::Tensor targets);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -77,4 +79,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_dgrad", &gemm_dgrad);
   m.def("gemm_wgrad", &gemm_wgrad);
   m.def("ce_bwd", &ce_bwd);
+  m.def("ce_partial_fwd", &ce_partial_fwd);
 }

@@ -127,25 +127,35 @@ class _VocabParallelCE(torch.autograd.Function):
                 vocab_start: int, vocab_end: int):
         group = get_tp_group()
         tp = get_tp_size()
-        x = logits.float()
-        N = x.shape[0]
-        local_max = x.max(dim=-1).values                      # (N,)
-        lse_local = torch.log(
-            torch.exp(x - local_max.unsqueeze(-1)).sum(-1)) + local_max
+        N = logits.shape[0]
+        fused = logits.is_cuda and logits.dtype == torch.bfloat16
+        if fused:
+            # ONE online pass over the (N, V/tp) shard (ops ce_partial_fwd):
+            # no fp32 logits materialization, no separate max/exp passes
+            from ...ops import ext
+            local_t = torch.where(
+                (target >= vocab_start) & (target < vocab_end),
+                target - vocab_start, torch.full_like(target, -1))
+            lse_local, tgt_logit = ext("ce_partial").ce_partial_fwd(
+                logits, local_t)
+        else:
+            x = logits.float()
+            local_max = x.max(dim=-1).values                  # (N,)
+            lse_local = torch.log(
+                torch.exp(x - local_max.unsqueeze(-1)).sum(-1)) + local_max
+            in_range = (target >= vocab_start) & (target < vocab_end)
+            local_t = (target - vocab_start).masked_fill(~in_range, 0)
+            tgt_logit = x.gather(-1, local_t.unsqueeze(-1)).squeeze(-1)
+            tgt_logit = tgt_logit.masked_fill(~in_range, 0.0)
         if tp > 1:
             all_lse = torch.empty(tp * N, dtype=lse_local.dtype,
                                   device=lse_local.device)
             dist.all_gather_into_tensor(all_lse, lse_local.contiguous(),
                                         group=group)
             lse = torch.logsumexp(all_lse.reshape(tp, N), dim=0)
+            dist.all_reduce(tgt_logit, op=dist.ReduceOp.SUM, group=group)
         else:
             lse = lse_local
-        in_range = (target >= vocab_start) & (target < vocab_end)
-        local_t = (target - vocab_start).masked_fill(~in_range, 0)
-        tgt_logit = x.gather(-1, local_t.unsqueeze(-1)).squeeze(-1)
-        tgt_logit = tgt_logit.masked_fill(~in_range, 0.0)
-        if tp > 1:
-            dist.all_reduce(tgt_logit, op=dist.ReduceOp.SUM, group=group)
         loss = (lse - tgt_logit).mean()
         ctx.save_for_backward(logits, target, lse)
         ctx.vocab_start, ctx.vocab_end = vocab_start, vocab_end
@@ -155,6 +165,17 @@ class _VocabParallelCE(torch.autograd.Function):
     def backward(ctx, gout):
         logits, target, lse = ctx.saved_tensors
         N = logits.shape[0]
+        if logits.is_cuda and logits.dtype == torch.bfloat16:
+            # ce_bwd with the GLOBAL lse: out-of-shard targets (-1) never
+            # match the one-hot test, so the kernel is reused verbatim
+            from ...ops import ext
+            local_t = torch.where(
+                (target >= ctx.vocab_start) & (target < ctx.vocab_end),
+                target - ctx.vocab_start, torch.full_like(target, -1))
+            grad = ext("ce_partial").ce_bwd(
+                logits, local_t, lse.contiguous(),
+                gout.reshape(1).float().contiguous())
+            return grad, None, None, None
         grad = torch.exp(logits.float() - lse.unsqueeze(-1))
         in_range = (target >= ctx.vocab_start) & (target < ctx.vocab_end)
         local_t = (target - ctx.vocab_start).masked_fill(~in_range, 0)
