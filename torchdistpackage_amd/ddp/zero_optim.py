@@ -195,14 +195,20 @@ class Bf16ZeroOptimizer:
     @torch.no_grad()
     def _finish_reduction(self):
         if self._grad_world() == 1:
-            # single-rank: no comm; master grads come straight from p.grad
+            # single-rank: no comm; master grads come straight from p.grad.
+            # Batched (ONE foreach cast-copy): the per-param loop was ~300
+            # bf16->fp32 kernel launches per step on Llama-8B (r02 profile)
+            dsts, srcs = [], []
             for i in self._my_idx:
                 p = self._params[i]
                 if p.grad is not None:
                     mp = self._master_params[i]
                     if mp.grad is None:
                         mp.grad = torch.empty_like(mp)
-                    mp.grad.copy_(p.grad)
+                    dsts.append(mp.grad)
+                    srcs.append(p.grad)
+            if dsts:
+                torch._foreach_copy_(dsts, srcs)
             return
         if self._use_gpu:
             cur = torch.cuda.current_stream()
@@ -259,6 +265,7 @@ class Bf16ZeroOptimizer:
         # copy (cast) owned grads into master grads; optionally free the rest.
         # Only params whose hook pushed THIS iteration: a never-fired param's
         # view holds a previous iteration's reduced grad (stale).
+        dsts, srcs = [], []
         for b in self._buckets:
             for p, v, pu in zip(b.params, b.views, b.pushed):
                 if not pu:
@@ -268,10 +275,13 @@ class Bf16ZeroOptimizer:
                     mp = self._master_params[i]
                     if mp.grad is None:
                         mp.grad = torch.empty_like(mp)
-                    mp.grad.copy_(v.view(p.shape))
+                    dsts.append(mp.grad)
+                    srcs.append(v.view(p.shape))
                 if self.stage2:
                     p.grad = None
             b.clear_pushed()
+        if dsts:
+            torch._foreach_copy_(dsts, srcs)   # one batched cast-copy
 
     # ------------------------------------------------------------------
 
@@ -306,11 +316,13 @@ class Bf16ZeroOptimizer:
     def _sync_params(self):
         """Copy updated master shard back into bf16 params and all-gather the
         full parameter set (ONE padded flat collective)."""
-        # local copy master -> model param for owned params
-        for i in self._my_idx:
-            self._params[i].copy_(
-                self._master_views[i].view(self._params[i].shape)
-                .to(self._params[i].dtype))
+        # local copy master -> model param for owned params (ONE batched
+        # foreach cast-copy, no intermediate .to() materialization)
+        if self._my_idx:
+            torch._foreach_copy_(
+                [self._params[i] for i in self._my_idx],
+                [self._master_views[i].view(self._params[i].shape)
+                 for i in self._my_idx])
         if not dist.is_initialized() or self.world == 1:
             return
         dtype = self._params[0].dtype

@@ -77,7 +77,9 @@ class ShardedEMA:
         for i in self._my_idx:
             p = self._params[i].detach().reshape(-1)
             v = self._views[i]
-            if p.dtype == torch.float32 and v.is_cuda:
+            if v.is_cuda and p.dtype in (torch.float32, torch.bfloat16):
+                # fused HIP pass (reads bf16 params directly — the lerp_
+                # fallback materialized p.float() every step)
                 from ..ops import ema_update_
                 ema_update_(v, p, d)
             else:
