@@ -578,7 +578,13 @@ namespace {
 // coalesced (each 32-lane half writes one q-row, 32 consecutive keys).  The
 // dq-lite kernel then consumes dS directly instead of recomputing
 // S/P/dP per tile (drops 32 of dq's 48 MFMAs plus its V/dO staging).
-template <bool CAUSAL, bool DK_PASS, bool WRITE_DS>
+// PMODE: 0 = self-contained; 1 = dv pass STORES its P register image
+// (bf16, per-lane-contiguous 32 B) to p_ws; 2 = dk pass LOADS that image
+// (2 x b128, issued before its dP MFMA chain) and SKIPS the 8-MFMA
+// S-recompute per q-subtile.  r01's P-store attempt lost 2x to 16 scalar
+// per-lane loads; the register-image layout makes the reload two vector
+// loads with T14-style early issue.
+template <bool CAUSAL, bool DK_PASS, bool WRITE_DS, int PMODE>
 __launch_bounds__(512)
 __global__ void attn_bwd_dkdv_v2_kernel(
     const unsigned short* __restrict__ q,
@@ -589,6 +595,7 @@ __global__ void attn_bwd_dkdv_v2_kernel(
     const float* __restrict__ delta,
     unsigned short* __restrict__ out,   // dk (DK_PASS) or dv
     unsigned short* __restrict__ ds_out, long spad,
+    unsigned short* __restrict__ p_ws,
     Strides2 qs, Strides2 ks, Strides2 vs, Strides2 dos, Strides2 outs,
     int B, int H, int S, float scale, int q_per_kv) {
   // LDS: Q and dO tiles (A-operands), plus the transposed tile the pass's
@@ -710,15 +717,32 @@ __global__ void attn_bwd_dkdv_v2_kernel(
 
 #pragma unroll
     for (int qt = 0; qt < 2; ++qt) {
+      // P register-image workspace slot for this (block, q-subtile, lane)
+      // q-subtile slots run over the PADDED q range (gridDim.x * 8
+      // subtiles of 32): with S not a multiple of 256, (S>>5) under-counts
+      // and neighbouring (bh, qsub) slots collide
+      unsigned short* pw = (PMODE != 0)
+          ? p_ws + (((((long)bh * ((long)gridDim.x * 8) +
+                       ((qt0 >> 5) + qt)) *
+                      gridDim.x + blockIdx.x) * 512 + tid) * 16)
+          : nullptr;
+      bf16x8 p_img[2];
+      if (PMODE == 2) {
+        // issue the P reload FIRST: its latency hides under the dP MFMAs
+        p_img[0] = *(const bf16x8*)pw;
+        p_img[1] = *(const bf16x8*)(pw + 8);
+      }
       f32x16 s_acc = (f32x16)(0.f);
       f32x16 dp_acc;
       if (DK_PASS) dp_acc = (f32x16)(0.f);
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        bf16x8_v a_q = pack8v(
-            &q_lds[buf][swzK(qt * 32 + l31, c * 16 + hi * 8)]);
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            a_q, kt_frag[c], s_acc, 0, 0, 0);
+        if (PMODE != 2) {
+          bf16x8_v a_q = pack8v(
+              &q_lds[buf][swzK(qt * 32 + l31, c * 16 + hi * 8)]);
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a_q, kt_frag[c], s_acc, 0, 0, 0);
+        }
         if (DK_PASS) {
           bf16x8_v a_do = pack8v(
               &do_lds[buf * KV * D2 + swzK(qt * 32 + l31, c * 16 + hi * 8)]);
@@ -732,13 +756,25 @@ __global__ void attn_bwd_dkdv_v2_kernel(
       for (int r = 0; r < 16; ++r) {
         int qrel = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
         int qrow = qt0 + qrel;
-        bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
-        float p = valid ?
-            __expf(s_acc[r] * scale - lse_lds[buf][qrel]) : 0.f;
+        float p;
+        if (PMODE == 2) {
+          p = bf2f(((const unsigned short*)p_img)[r]);
+        } else {
+          bool valid = key < S && qrow < S && (!CAUSAL || key <= qrow);
+          p = valid ?
+              __expf(s_acc[r] * scale - lse_lds[buf][qrel]) : 0.f;
+        }
         s_acc[r] = DK_PASS
             ? p * (dp_acc[r] - del_lds[buf][qrel]) * scale : p;
         if (DK_PASS && WRITE_DS && qrow < S)
           ds_out[((long)bh * S + qrow) * spad + key] = f2bf(s_acc[r]);
+      }
+      if (PMODE == 1) {
+        unsigned short pb[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) pb[r] = f2bf(s_acc[r]);
+        *(bf16x8*)pw = *(const bf16x8*)pb;
+        *(bf16x8*)(pw + 8) = *(const bf16x8*)(pb + 8);
       }
 
       // B-fragments over the q dimension (exchange as in the forward)
@@ -815,31 +851,33 @@ void attn_bwd_dkdv_v2(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   };
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid((S + QT2 - 1) / QT2, B * H), block(512);
-#define L_KV2(CC, WDS, DSP, SPAD)                                             \
+#define L_KV2(CC, WDS, DSP, SPAD, PW, PM1, PM2)                               \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false, false>), grid,     \
-                       block, 0,                                              \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, false, false, PM1>),      \
+                       grid, block, 0,                                        \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
                        (const unsigned short*)v.data_ptr(),                   \
                        (const unsigned short*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (unsigned short*)dv.data_ptr(), nullptr, 0, get(q),    \
+                       (unsigned short*)dv.data_ptr(), nullptr, 0, PW,        \
+                       get(q),                                                \
                        get(k), get(v), get(dout), get(dv), B, H, S,           \
                        (float)scale, q_per_kv);                               \
-    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true, WDS>), grid,        \
-                       block, 0,                                              \
+    hipLaunchKernelGGL((attn_bwd_dkdv_v2_kernel<CC, true, WDS, PM2>),         \
+                       grid, block, 0,                                        \
                        stream, (const unsigned short*)q.data_ptr(),           \
                        (const unsigned short*)k.data_ptr(),                   \
                        (const unsigned short*)v.data_ptr(),                   \
                        (const unsigned short*)dout.data_ptr(),                \
                        lse.data_ptr<float>(), delta.data_ptr<float>(),        \
-                       (unsigned short*)dk.data_ptr(), DSP, SPAD, get(q),     \
+                       (unsigned short*)dk.data_ptr(), DSP, SPAD, PW,         \
+                       get(q),                                                \
                        get(k), get(v), get(dout), get(dk), B, H, S,           \
                        (float)scale, q_per_kv);                               \
   } while (0)
-  if (causal) L_KV2(true, false, nullptr, 0);
-  else L_KV2(false, false, nullptr, 0);
+  if (causal) L_KV2(true, false, nullptr, 0, nullptr, 0, 0);
+  else L_KV2(false, false, nullptr, 0, nullptr, 0, 0);
   HIP_CHECK_LAST();
 }
 
@@ -1127,7 +1165,8 @@ void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   {
     const long Sq = q.size(2);
     const long spad_est = (Sq + QT2 - 1) / QT2 * QT2;
-    const long ds_bytes = (long)q.size(0) * q.size(1) * Sq * spad_est * 2;
+    const long pmul = std::getenv("TDPA_PSTORE") != nullptr ? 4 : 2;
+    const long ds_bytes = (long)q.size(0) * q.size(1) * Sq * spad_est * pmul;
     if (recompute || ds_bytes > kMaxDsBytes) {
       attn_bwd_dq_v2(dout, q, k, v, lse, delta, dq, causal, scale);
       attn_bwd_dkdv_v2(dout, q, k, v, lse, delta, dk, dv, causal, scale);
@@ -1146,8 +1185,26 @@ void attn_bwd_v2_all(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   const long spad = (long)grid.x * QT2;
   auto dsw = torch::empty({(long)B * H, (long)S, spad}, q.options());
   unsigned short* dsp = (unsigned short*)dsw.data_ptr();
-  if (causal) L_KV2(true, true, dsp, spad);
-  else L_KV2(false, true, dsp, spad);
+  // P register-image workspace (opt-in, TDPA_PSTORE): the dv pass stores
+  // P, the dk pass reloads it (2 early b128/lane) and skips its 8-MFMA S
+  // recompute.  Measured NEUTRAL at the bench shape (bwd 891-909 us both
+  // ways): the saved MFMAs are repaid by the extra S^2 bf16 write+read —
+  // kept for shapes where the balance may differ, default OFF so the
+  // backward's workspace stays a single dS buffer.
+  static const bool pstore = std::getenv("TDPA_PSTORE") != nullptr;
+  torch::Tensor pwt;
+  unsigned short* pw = nullptr;
+  if (pstore) {
+    pwt = torch::empty({(long)B * H, spad, spad}, q.options());
+    pw = (unsigned short*)pwt.data_ptr();
+  }
+  if (causal) {
+    if (pw) L_KV2(true, true, dsp, spad, pw, 1, 2);
+    else L_KV2(true, true, dsp, spad, nullptr, 0, 0);
+  } else {
+    if (pw) L_KV2(false, true, dsp, spad, pw, 1, 2);
+    else L_KV2(false, true, dsp, spad, nullptr, 0, 0);
+  }
 #undef L_KV2
   static const bool no_tr16 = std::getenv("TDPA_NO_TR16") != nullptr;
 #define L_DQL(CC)                                                             \
