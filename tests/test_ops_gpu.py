@@ -564,3 +564,35 @@ def test_ce_partial_fwd_kernel():
                                            .unsqueeze(-1)).squeeze(-1),
                           torch.zeros_like(ref_lse))
     assert (tgt - ref_tgt).abs().max() < 1e-3
+
+
+def test_gqa_attention_zero_copy():
+    """Llama zero-copy attention core (strided v in, (S,B,HD) out) vs the
+    fp32 oracle with repeated KV heads."""
+    from torchdistpackage_amd.ops import gqa_attention
+    torch.manual_seed(9)
+    B, H, Hkv, S, D = 2, 8, 2, 256, 128
+    qb = (torch.randn(S, B, H * D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    kb = (torch.randn(B, Hkv, S, D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    vb = (torch.randn(S, B, Hkv * D) * 0.3).bfloat16().to(_dev()).requires_grad_(True)
+    q4 = qb.view(S, B, H, D).permute(1, 2, 0, 3).contiguous()
+    v4 = vb.view(S, B, Hkv, D).permute(1, 2, 0, 3)   # STRIDED view
+    o = gqa_attention(q4, kb, v4, causal=True)
+    assert o.shape == (S, B, H * D)
+
+    qf = q4.detach().float().requires_grad_(True)
+    kf = kb.detach().float().requires_grad_(True)
+    vf = v4.detach().float().requires_grad_(True)
+    kr = kf.repeat_interleave(H // Hkv, 1)
+    vr = vf.repeat_interleave(H // Hkv, 1)
+    s = qf @ kr.transpose(-1, -2) / (D ** 0.5)
+    mask = torch.ones(S, S, dtype=torch.bool, device=_dev()).tril_()
+    ref4 = torch.softmax(s.masked_fill(~mask, float("-inf")), -1) @ vr
+    ref = ref4.permute(2, 0, 1, 3).reshape(S, B, H * D)
+    assert (o.float() - ref).abs().max() < 3e-2
+    g = torch.randn_like(ref)
+    o.backward(g.bfloat16())
+    ref.backward(g)
+    gq = qb.grad.view(S, B, H, D).permute(1, 2, 0, 3)
+    assert (gq.float() - qf.grad).abs().max() < 6e-2
+    assert (kb.grad.float() - kf.grad).abs().max() < 6e-2

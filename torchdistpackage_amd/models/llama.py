@@ -110,15 +110,23 @@ class LlamaAttention(nn.Module):
         k = fast_linear(x, self.wk.weight)
         v = fast_linear(x, self.wv.weight)
 
-        def shape(t, nh):
-            return t.reshape(S, B, nh, self.hd).permute(1, 2, 0, 3).contiguous()
+        # zero-copy layouts: RoPE reads the permuted projection views in
+        # place (strided kernel input); V goes to the flash kernel as a
+        # strided view; the attention output is written straight into an
+        # (S, B, H*hd) buffer for the out-projection GEMM.  The permuted
+        # .contiguous() chain this replaces cost ~4 large copies per layer.
+        def view4(t, nh):
+            return t.reshape(S, B, nh, self.hd).permute(1, 2, 0, 3)
 
-        q = self.rope(shape(q, self.nh_local))
-        k = self.rope(shape(k, self.nkv_local))
-        v = shape(v, self.nkv_local)
-        # GQA handled natively by the flash kernels (kv head = q head // rep)
-        o = flash_attention(q, k, v, causal=self.causal)
-        o = o.permute(2, 0, 1, 3).reshape(S, B, self.nh_local * self.hd)
+        q = self.rope(view4(q, self.nh_local))      # -> contiguous (B,H,S,D)
+        k = self.rope(view4(k, self.nkv_local))
+        v = view4(v, self.nkv_local)
+        if q.is_cuda and q.dtype == torch.bfloat16 and self.hd in (64, 128):
+            from ..ops import gqa_attention
+            o = gqa_attention(q, k, v, causal=self.causal)
+        else:
+            o = flash_attention(q, k, v.contiguous(), causal=self.causal)
+            o = o.permute(2, 0, 1, 3).reshape(S, B, self.nh_local * self.hd)
         return self.wo(o)
 
 

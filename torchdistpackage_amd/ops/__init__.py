@@ -469,6 +469,62 @@ def fused_qkv_attention(qkv, n_head: int, causal: bool = True,
     return _FusedQKVAttentionFn.apply(qkv, n_head, causal, scale)
 
 
+class _GqaAttentionFn(torch.autograd.Function):
+    """Flash attention for the Llama path with zero-copy layouts.
+
+    q (B,H,S,D) contiguous (RoPE output); k (B,Hkv,S,D) contiguous; v may be
+    a STRIDED view of the (S,B,Hkv*hd) projection output.  Returns
+    (S, B, H*hd) — the kernels write the output through a strided view, so
+    no permute-contiguous copy feeds the out-projection GEMM.  dq is
+    likewise written straight into an (S,B,H*hd)-backed view; GQA dk/dv
+    partials (q-head width) are summed to kv width.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        B, H, S, D = q.shape
+        o_buf = torch.empty(S, B, H * D, dtype=q.dtype, device=q.device)
+        o4 = o_buf.view(S, B, H, D).permute(1, 2, 0, 3)
+        _, lse = ext("flash_attention").attn_fwd(
+            q, k, v, o4, bool(causal), float(scale))
+        ctx.save_for_backward(q, k, v, o_buf, lse)
+        ctx.meta = (causal, scale)
+        return o_buf
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o_buf, lse = ctx.saved_tensors
+        causal, scale = ctx.meta
+        B, H, S, D = q.shape
+        Hkv = k.shape[1]
+        rep = H // Hkv
+        o4 = o_buf.view(S, B, H, D).permute(1, 2, 0, 3)
+        do = do.contiguous()
+        do4 = do.view(S, B, H, D).permute(1, 2, 0, 3)
+        dq_buf = torch.empty(S, B, H * D, dtype=q.dtype, device=q.device)
+        dq = dq_buf.view(S, B, H, D).permute(1, 2, 0, 3)
+        dk = torch.empty(B, H, S, D, dtype=k.dtype, device=k.device)
+        dv = torch.empty(B, H, S, D, dtype=v.dtype, device=v.device)
+        ext("flash_attention").attn_bwd(do4, q, k, v, o4, lse, dq, dk, dv,
+                                        causal, scale)
+        if rep > 1:
+            dk = dk.view(B, Hkv, rep, S, D).sum(2)
+            dv = dv.view(B, Hkv, rep, S, D).sum(2)
+        return dq, dk, dv, None, None
+
+
+def gqa_attention(q, k, v, causal: bool = True, scale: float = None):
+    """Zero-copy GQA flash attention (GPU bf16 path; see _GqaAttentionFn).
+    Falls back to :func:`flash_attention` off-GPU / off-bf16."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+        return _GqaAttentionFn.apply(q, k, v, causal, scale)
+    o = flash_attention(q, k, v, causal=causal, scale=scale)
+    B, H, S, D = q.shape
+    return o.permute(2, 0, 1, 3).reshape(S, B, H * D)
+
+
 class _CrossEntropyFn(torch.autograd.Function):
     """Fused mean cross-entropy on bf16 logits (one online-LSE pass fwd, one
     dlogits pass bwd) — replaces logits.float()+F.cross_entropy which
@@ -510,8 +566,9 @@ class _RopeFn(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         ctx.pos0 = pos0
         if x.is_cuda and x.dtype == torch.bfloat16:
-            return ext("rope").rope_apply(x.contiguous(), cos, sin, pos0,
-                                          True)
+            # strided input is read in place (head dim must be contiguous);
+            # output is contiguous (B, H, S, D)
+            return ext("rope").rope_apply(x, cos, sin, pos0, True)
         hd = x.shape[-1]
         S = x.shape[-2]
         c = cos[pos0:pos0 + S].to(x.dtype)
@@ -523,7 +580,9 @@ class _RopeFn(torch.autograd.Function):
     def backward(ctx, dy):
         cos, sin = ctx.saved_tensors
         if dy.is_cuda and dy.dtype == torch.bfloat16:
-            return (ext("rope").rope_apply(dy.contiguous(), cos, sin,
+            if dy.stride(-1) != 1:
+                dy = dy.contiguous()
+            return (ext("rope").rope_apply(dy, cos, sin,
                                            ctx.pos0, False),
                     None, None, None)
         hd = dy.shape[-1]

@@ -34,13 +34,18 @@ DEVINL void store8(unsigned short* p, const float* f) {
 }
 
 // rows = B*H*S (seq position = row % S), rotate-half pairs (i, i+D/2).
-// FWD=false computes the transpose rotation (gradient).
+// FWD=false computes the transpose rotation (gradient).  The INPUT may be
+// any-strided over (B,H,S) with d contiguous (e.g. a permuted view of the
+// (S,B,H*hd) projection output — reading it in place removes the
+// permute-contiguous copy per q/k per layer); the output is written
+// contiguous (B,H,S,D).
 template <bool FWD>
 __global__ void rope_kernel(const unsigned short* __restrict__ x,
                             unsigned short* __restrict__ y,
                             const float* __restrict__ cs,   // (S, D/2) cos
                             const float* __restrict__ sn,   // (S, D/2) sin
-                            long rows, int S, int D, int pos0) {
+                            long rows, int S, int D, int pos0,
+                            long H, long xsb, long xsh, long xss) {
   const int half = D >> 1;
   const int chunks = half >> 3;                 // 8 elems per thread-chunk
   const long total = rows * chunks;
@@ -49,10 +54,13 @@ __global__ void rope_kernel(const unsigned short* __restrict__ x,
     const long row = g / chunks;
     const int c = (int)(g - row * chunks) * 8;
     const int s = (int)(row % S) + pos0;
+    const long bh = row / S;
+    const long xbase = (bh / H) * xsb + (bh - (bh / H) * H) * xsh +
+                       (row % S) * xss + c;
     const long base = row * D + c;
     float x1[8], x2[8], co[8], si[8];
-    load8(x + base, x1);
-    load8(x + base + half, x2);
+    load8(x + xbase, x1);
+    load8(x + xbase + half, x2);
 #pragma unroll
     for (int j = 0; j < 8; j += 4) {
       *(float4*)(co + j) = *(const float4*)(cs + (long)s * half + c + j);
@@ -128,13 +136,16 @@ void check_bf16(const torch::Tensor& t, const char* n) {
 // x (B, H, S, D) bf16 contiguous; cos/sin (>=S+pos0, D/2) fp32.
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
                          long pos0, bool fwd) {
-  check_bf16(x, "x");
-  TORCH_CHECK(x.dim() == 4 && cs.scalar_type() == torch::kFloat &&
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+              x.dim() == 4 && x.stride(3) == 1,
+              "rope: x must be 4-D bf16 CUDA with contiguous head dim");
+  TORCH_CHECK(cs.scalar_type() == torch::kFloat &&
               cs.is_contiguous() && sn.is_contiguous());
   const int D = x.size(3), S = x.size(2);
   TORCH_CHECK(D % 16 == 0 && cs.size(1) == D / 2);
   const long rows = (long)x.size(0) * x.size(1) * S;
-  auto y = torch::empty_like(x);
+  auto y = torch::empty({x.size(0), x.size(1), x.size(2), x.size(3)},
+                        x.options());
   const long total = rows * (D / 16);
   const long nb = (total + BLOCK - 1) / BLOCK;
   auto stream = at::cuda::getCurrentHIPStream();
@@ -143,7 +154,8 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
                      dim3(BLOCK), 0, stream,
                      (const unsigned short*)x.data_ptr(),
                      (unsigned short*)y.data_ptr(), cs.data_ptr<float>(),
-                     sn.data_ptr<float>(), rows, S, D, (int)pos0);
+                     sn.data_ptr<float>(), rows, S, D, (int)pos0,
+                     x.size(1), x.stride(0), x.stride(1), x.stride(2));
   HIP_CHECK_LAST();
   return y;
 }
