@@ -233,3 +233,69 @@ def test_1f1b_extra_inputs_pp2():
 
 def test_1f1b_extra_inputs_pp3():
     run_distributed(_pp_extra_inputs, world_size=3)
+
+
+def _pp_tp_vocab_parallel_parity(rank, world_size):
+    """The 8-GPU flagship composition in miniature (pp2 x tp2, gloo):
+    GPT-2 stages with the vocab-parallel embedding/head/CE across PP
+    boundaries must reproduce the tp=1/pp=1 oracle loss (weight surgery)."""
+    import torch.distributed as dist
+    from torchdistpackage_amd.dist.topo import tpc
+    from torchdistpackage_amd.parallel.tensor import set_tp_group
+    from torchdistpackage_amd.models.gpt2 import GPT2Config, GPT2Model
+    from torchdistpackage_amd.parallel.pipeline import (forward_backward,
+                                                        partition_uniform)
+    import sys, os
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from bench_pp import PPStage
+
+    cfg = GPT2Config(vocab_size=128, n_layer=2, n_head=4, dim=32, max_seq=16,
+                     tie_weights=False)
+    torch.manual_seed(0)
+    oracle = GPT2Model(cfg)       # before TP groups exist: tp=1 semantics
+    torch.manual_seed(2)
+    x = torch.randint(0, 128, (4, 16))
+    loss_ref = float(oracle(x, labels=x)["loss"])
+
+    tpc.setup_process_groups([("pipe", 2), ("tensor", 2)])
+    set_tp_group(tpc.get_group("tensor"))
+    torch.manual_seed(0)
+    tp_model = GPT2Model(cfg)
+    # surgery: oracle weights -> TP shards
+    tp_model.embed.wte.load_from_full(oracle.embed.wte.weight)
+    tp_model.embed.wpe.load_state_dict(oracle.embed.wpe.state_dict())
+    tp_model.head.ln_f.load_state_dict(oracle.head.ln_f.state_dict())
+    with torch.no_grad():
+        tp_model.head.weight.copy_(
+            oracle.head.weight[tp_model.head.vocab_start:
+                               tp_model.head.vocab_end])
+    for fb, tb in zip(oracle.blocks, tp_model.blocks):
+        tb.init_from_full(fb)
+
+    layers = tp_model.to_stage_layers()
+    parts = partition_uniform(len(layers), 2)
+    s, e = parts[tpc.get_pp_rank()]
+    num_mb = 2
+    stage = PPStage(layers[s:e], tpc.is_first_in_pipeline_group(),
+                    tpc.is_last_in_pipeline_group(), num_mb, 2)
+
+    def fwd_fn(stage_in, labels=None):
+        return stage(stage_in, labels=labels)
+
+    losses = forward_backward(
+        fwd_fn, inputs=x if stage.is_first else None,
+        num_microbatches=num_mb,
+        extra_inputs=x if stage.is_last else None,
+        return_losses=True)
+    if stage.is_last:
+        total = sum(float(l) for l in losses)
+        assert abs(total - loss_ref) < 2e-3, (total, loss_ref)
+    # every stage's params must have grads (except frozen none here)
+    for n, p in stage.named_parameters():
+        assert p.grad is not None, n
+    return True
+
+
+def test_pp2_tp2_vocab_parallel_parity():
+    run_distributed(_pp_tp_vocab_parallel_parity, world_size=4)
