@@ -173,3 +173,37 @@ def _zero_tp_sp_grads(rank, world_size):
 
 def test_zero_tp_sp_grads_world4():
     run_distributed(_zero_tp_sp_grads, world_size=4)
+
+
+def _zero_grad_acc(rank, world_size):
+    """num_grad_acc_iter: two accumulated backwards per step must match one
+    full-batch step (DDP+Adam oracle), with no intermediate reduces."""
+    from torchdistpackage_amd.ddp import Bf16ZeroOptimizer
+
+    model_a = _make_model(seed=21)
+    model_b = copy.deepcopy(model_a)
+    inner = torch.optim.Adam(model_a.parameters(), lr=1e-3)
+    zopt = Bf16ZeroOptimizer(inner, num_grad_acc_iter=2)
+    ref = nn.parallel.DistributedDataParallel(model_b)
+    opt_b = torch.optim.Adam(model_b.parameters(), lr=1e-3)
+
+    for it in range(3):
+        torch.manual_seed(900 + 10 * it + rank)
+        x = torch.randn(8, 48)
+        # accumulated halves (scaled so grads match the full batch)
+        (model_a(x[:4]).pow(2).mean() / 2).backward()
+        (model_a(x[4:]).pow(2).mean() / 2).backward()
+        (ref(x[:4]).pow(2).mean() / 2).backward()
+        (ref(x[4:]).pow(2).mean() / 2).backward()
+        zopt.step()
+        opt_b.step()
+        zopt.zero_grad()
+        opt_b.zero_grad()
+        for (na, pa), (nb, pb) in zip(model_a.named_parameters(),
+                                      model_b.named_parameters()):
+            assert torch.allclose(pa, pb, atol=1e-5), (it, na)
+    return True
+
+
+def test_zero_grad_accumulation():
+    run_distributed(_zero_grad_acc, world_size=2)

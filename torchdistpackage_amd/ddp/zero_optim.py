@@ -49,7 +49,8 @@ class Bf16ZeroOptimizer:
                  stage2: bool = False, overlap_comm: bool = True,
                  bucket_cap_mb: float = 100.0,
                  bf16_master_weights: bool = False,
-                 clip_grad: float = 0.0):
+                 clip_grad: float = 0.0,
+                 num_grad_acc_iter: int = 1):
         """
         Args:
             optimizer: inner optimizer already constructed over the model
@@ -60,6 +61,10 @@ class Bf16ZeroOptimizer:
                 pass the full DP group when ``group`` is a node group).
             stage2: free non-owned grads right after reduction (ZeRO-2).
             clip_grad: if >0, clip global grad norm before step.
+            num_grad_acc_iter: with gradient accumulation / PP
+                micro-batching, only every N-th backward communicates
+                (NaiveDdp parity; intermediate backwards just accumulate —
+                round-1 ZeRO re-reduced every micro-batch, wasted comm).
         """
         self.optim = optimizer
         self.group = group
@@ -68,6 +73,8 @@ class Bf16ZeroOptimizer:
         self.overlap = overlap_comm
         self.clip_grad = clip_grad
         self.bucket_cap = int(bucket_cap_mb * 1024 * 1024)
+        self.num_grad_acc_iter = max(1, num_grad_acc_iter)
+        self._fires: Dict[int, int] = {}
 
         self.rank = dist.get_rank(self.group) if dist.is_initialized() else 0
         self.world = dist.get_world_size(self.group) if dist.is_initialized() else 1
@@ -167,6 +174,11 @@ class Bf16ZeroOptimizer:
     def _on_grad_ready(self, p: torch.Tensor):
         if self._grad_world() == 1:
             return
+        if self.num_grad_acc_iter > 1:
+            fires = self._fires.get(id(p), 0) + 1
+            self._fires[id(p)] = fires
+            if fires % self.num_grad_acc_iter != 0:
+                return      # intermediate micro-batch: accumulate only
         bucket, idx = self._param_bucket[id(p)]
         if bucket.push(idx, p.grad):
             self._reduce_bucket(bucket)
