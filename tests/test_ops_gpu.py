@@ -596,3 +596,28 @@ def test_gqa_attention_zero_copy():
     gq = qb.grad.view(S, B, H, D).permute(1, 2, 0, 3)
     assert (gq.float() - qf.grad).abs().max() < 6e-2
     assert (kb.grad.float() - kf.grad).abs().max() < 6e-2
+
+
+def test_vocab_parallel_ce_fused_vs_eager():
+    """_VocabParallelCE's fused GPU kernels vs its own eager branch on
+    SHARD inputs (out-of-range targets included) — this path otherwise
+    first runs inside the driver's multi-GPU job."""
+    from torchdistpackage_amd.parallel.tensor.vocab import _VocabParallelCE
+    torch.manual_seed(11)
+    N, Vp, vs = 96, 256, 256   # shard covering vocab ids [256, 512)
+    logits0 = (torch.randn(N, Vp) * 2).bfloat16().to(_dev())
+    target = torch.randint(0, 768, (N,), device=_dev())  # many out of shard
+
+    l_fused = logits0.clone().requires_grad_(True)
+    loss_f = _VocabParallelCE.apply(l_fused, target, vs, vs + Vp)
+    loss_f.backward()
+
+    # eager branch: force the non-fused path with an fp32 clone
+    l_eager = logits0.clone().float().requires_grad_(True)
+    loss_e = _VocabParallelCE.apply(l_eager, target, vs, vs + Vp)
+    loss_e.backward()
+
+    assert abs(loss_f.item() - loss_e.item()) < 2e-3
+    ge = l_eager.grad
+    gf = l_fused.grad.float()
+    assert (gf - ge).abs().max() < 2e-3, (gf - ge).abs().max()
