@@ -263,3 +263,36 @@ def _moe_ep1_in_multirank_world(rank, world_size):
 
 def test_moe_ep1_guard_world2():
     run_distributed(_moe_ep1_in_multirank_world, world_size=2)
+
+
+def test_batched_experts_parity():
+    """BatchedExperts (two baddbmm over all local experts, clamp-gather
+    padding) vs the per-expert ModuleList path, identical weights."""
+    from torchdistpackage_amd.moe.layer import ExpertParallelMoE
+    torch.manual_seed(13)
+    dim = 32
+    a = ExpertParallelMoE(dim, num_experts=4, top_k=2, hidden_mult=2,
+                          ep_group=None, batched=False)
+    b = ExpertParallelMoE(dim, num_experts=4, top_k=2, hidden_mult=2,
+                          ep_group=None, batched=True)
+    b.load_state_dict({k: v for k, v in a.state_dict().items()
+                       if k.startswith("router")}, strict=False)
+    b.experts_b.load_from_experts(a.experts)
+    with torch.no_grad():
+        b.router.gate.weight.copy_(a.router.gate.weight)
+
+    x = torch.randn(6, 5, dim, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    ya = a(x)
+    yb = b(x2)
+    assert torch.allclose(ya, yb, atol=1e-5), (ya - yb).abs().max()
+    g = torch.randn_like(ya)
+    ya.backward(g)
+    yb.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+    # expert grads match: fc1 of expert e <-> w1[e]
+    for e, ex in enumerate(a.experts):
+        assert torch.allclose(b.experts_b.w1.grad[e], ex.fc1.weight.grad,
+                              atol=1e-5), e
+        assert torch.allclose(b.experts_b.b2.grad[e], ex.fc2.bias.grad,
+                              atol=1e-5), e

@@ -48,7 +48,7 @@ class MoEBlock(nn.Module):
         self.attn = Attention(cfg.dim, cfg.n_head, causal=cfg.causal, **kw)
         self.ln_2 = LayerNorm(cfg.dim, **kw)
         self.moe = ExpertParallelMoE(cfg.dim, cfg.num_experts, cfg.top_k,
-                                     cfg.hidden_mult, **kw)
+                                     cfg.hidden_mult, batched=True, **kw)
 
     def forward(self, x):
         x = x + self.attn(self.ln_1(x))

@@ -89,6 +89,65 @@ class Expert(nn.Module):
         return fast_linear(h, self.fc2.weight, self.fc2.bias)
 
 
+class BatchedExperts(nn.Module):
+    """All local experts as stacked (E, ...) parameters, run as TWO
+    baddbmm calls over every expert at once.
+
+    The per-expert nn.Linear loop (r02 MoE profile) spent ~19% of the step
+    in its padding fills, cat/scatter copies and per-expert GEMM launches
+    (946 expert GEMMs + 2096 fills + 2779 copies per 3 steps).  Here the
+    per-expert token segments are CLAMP-GATHERED into an (E, maxn, D)
+    batch — padding rows replicate a real row instead of zero-fill, their
+    outputs are never gathered back, so they contribute exactly zero
+    gradient — and the whole MoE FFN is baddbmm -> fused GELU -> baddbmm.
+    """
+
+    def __init__(self, num_local: int, dim: int, hidden_mult: int = 4,
+                 device=None, dtype=None):
+        super().__init__()
+        kw = {"device": device, "dtype": dtype}
+        H = dim * hidden_mult
+        self.w1 = nn.Parameter(torch.empty(num_local, H, dim, **kw))
+        self.b1 = nn.Parameter(torch.zeros(num_local, H, **kw))
+        self.w2 = nn.Parameter(torch.empty(num_local, dim, H, **kw))
+        self.b2 = nn.Parameter(torch.zeros(num_local, dim, **kw))
+        nn.init.normal_(self.w1, std=0.02)
+        nn.init.normal_(self.w2, std=0.02)
+
+    @torch.no_grad()
+    def load_from_experts(self, experts):
+        for e, ex in enumerate(experts):
+            self.w1[e].copy_(ex.fc1.weight)
+            self.b1[e].copy_(ex.fc1.bias)
+            self.w2[e].copy_(ex.fc2.weight)
+            self.b2[e].copy_(ex.fc2.bias)
+
+    def forward(self, grouped: torch.Tensor, cnt: torch.Tensor,
+                maxn: int) -> torch.Tensor:
+        """grouped (N, D) tokens sorted by local expert; cnt (E,) device
+        counts; returns (N, D) in the same order."""
+        E = self.w1.shape[0]
+        D = grouped.shape[1]
+        if grouped.shape[0] == 0:
+            return grouped
+        # quantize the batch width so the hipBLASLt heuristic shape set
+        # stays tiny (routing drifts every step — r01 v7 lesson)
+        Q = 512
+        maxn_pad = max((maxn + Q - 1) // Q * Q, Q)
+        offs = torch.cumsum(cnt, 0) - cnt
+        ar = torch.arange(maxn_pad, device=grouped.device)
+        idx = offs[:, None] + torch.minimum(
+            ar[None, :], (cnt[:, None] - 1).clamp(min=0))
+        xg = grouped.index_select(0, idx.reshape(-1)).view(E, maxn_pad, D)
+        h = torch.baddbmm(self.b1.unsqueeze(1), xg,
+                          self.w1.transpose(1, 2))
+        h = bias_gelu(h, None)
+        y = torch.baddbmm(self.b2.unsqueeze(1), h,
+                          self.w2.transpose(1, 2))
+        valid = ar[None, :] < cnt[:, None]
+        return y.reshape(-1, D)[valid.reshape(-1)]
+
+
 def _all_to_all_uneven(x: torch.Tensor, in_splits: List[int],
                        out_splits: List[int],
                        group: Optional[dist.ProcessGroup]) -> torch.Tensor:
@@ -149,6 +208,7 @@ class ExpertParallelMoE(nn.Module):
     def __init__(self, dim: int, num_experts: int, top_k: int = 2,
                  hidden_mult: int = 4,
                  ep_group: Optional[dist.ProcessGroup] = None,
+                 batched: bool = False,
                  device=None, dtype=None):
         super().__init__()
         if ep_group is None:
@@ -169,10 +229,16 @@ class ExpertParallelMoE(nn.Module):
         self.top_k = top_k
         self.router = TopKRouter(dim, num_experts, top_k, device=device,
                                  dtype=dtype)
-        self.experts = nn.ModuleList([
-            Expert(dim, hidden_mult, device=device, dtype=dtype)
-            for _ in range(self.num_local)])
-        mark_expert_parallel(self.experts)
+        self.batched = batched
+        if batched:
+            self.experts_b = BatchedExperts(self.num_local, dim, hidden_mult,
+                                            device=device, dtype=dtype)
+            mark_expert_parallel(self.experts_b)
+        else:
+            self.experts = nn.ModuleList([
+                Expert(dim, hidden_mult, device=device, dtype=dtype)
+                for _ in range(self.num_local)])
+            mark_expert_parallel(self.experts)
         self.aux_loss = torch.zeros(())  # last forward's aux loss
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -243,21 +309,26 @@ class ExpertParallelMoE(nn.Module):
         # every UNSEEN (M,N,K) costs a host-side hipBLASLt heuristic pass
         # (~5-10 ms): unquantized expert GEMMs made deep MoE forwards
         # host-bound (measured 90 ms/block cold vs 2 ms warm).
-        Q = 1024
-        y_parts = []
-        off = 0
-        for le in range(self.num_local):
-            n = per_expert[le]
-            if n > 0:
-                seg = grouped[off:off + n]
-                npad = (n + Q - 1) // Q * Q
-                if npad != n:
-                    pad = torch.zeros(npad - n, seg.shape[1],
-                                      dtype=seg.dtype, device=seg.device)
-                    seg = torch.cat([seg, pad], dim=0)
-                y_parts.append(self.experts[le](seg)[:n])
-            off += n
-        y_all = torch.cat(y_parts, dim=0) if y_parts else grouped[:0]
+        if self.batched:
+            cnt_dev = my_slice.sum(0).to(received.device)
+            y_all = self.experts_b(grouped, cnt_dev,
+                                   max(per_expert) if per_expert else 0)
+        else:
+            Q = 1024
+            y_parts = []
+            off = 0
+            for le in range(self.num_local):
+                n = per_expert[le]
+                if n > 0:
+                    seg = grouped[off:off + n]
+                    npad = (n + Q - 1) // Q * Q
+                    if npad != n:
+                        pad = torch.zeros(npad - n, seg.shape[1],
+                                          dtype=seg.dtype, device=seg.device)
+                        seg = torch.cat([seg, pad], dim=0)
+                    y_parts.append(self.experts[le](seg)[:n])
+                off += n
+            y_all = torch.cat(y_parts, dim=0) if y_parts else grouped[:0]
         outs = torch.empty_like(received)
         outs[order2] = y_all
 
