@@ -297,3 +297,20 @@ def test_batched_experts_parity():
                               atol=1e-5), e
         assert torch.allclose(b.experts_b.b2.grad[e], ex.fc2.bias.grad,
                               atol=1e-5), e
+
+
+def test_batched_experts_empty_expert():
+    """Empty experts (routing collapse) must not index past the token
+    buffer — the bench crashed on exactly this before the clamp."""
+    from torchdistpackage_amd.moe.layer import BatchedExperts
+    torch.manual_seed(17)
+    be = BatchedExperts(num_local=4, dim=16, hidden_mult=2)
+    grouped = torch.randn(10, 16)
+    # experts 2 and 3 (the LAST) empty
+    cnt = torch.tensor([4, 6, 0, 0])
+    y = be(grouped, cnt, 6)
+    assert y.shape == (10, 16)
+    assert torch.isfinite(y).all()
+    # all-empty tail + zero tokens entirely
+    y2 = be(grouped[:0], torch.zeros(4, dtype=torch.long), 0)
+    assert y2.shape == (0, 16)

@@ -138,6 +138,10 @@ class BatchedExperts(nn.Module):
         ar = torch.arange(maxn_pad, device=grouped.device)
         idx = offs[:, None] + torch.minimum(
             ar[None, :], (cnt[:, None] - 1).clamp(min=0))
+        # an EMPTY expert whose offset sits at the end of `grouped` would
+        # index one past the buffer (routing collapse mid-training did
+        # exactly this); padding rows are never read back, any row works
+        idx = idx.clamp_(0, grouped.shape[0] - 1)
         xg = grouped.index_select(0, idx.reshape(-1)).view(E, maxn_pad, D)
         h = torch.baddbmm(self.b1.unsqueeze(1), xg,
                           self.w1.transpose(1, 2))
