@@ -143,11 +143,12 @@ class BatchedExperts(nn.Module):
         # exactly this); padding rows are never read back, any row works
         idx = idx.clamp_(0, grouped.shape[0] - 1)
         xg = grouped.index_select(0, idx.reshape(-1)).view(E, maxn_pad, D)
-        h = torch.baddbmm(self.b1.unsqueeze(1), xg,
-                          self.w1.transpose(1, 2))
-        h = bias_gelu(h, None)
-        y = torch.baddbmm(self.b2.unsqueeze(1), h,
-                          self.w2.transpose(1, 2))
+        # bmm + fused per-expert bias+GELU: hipBLASLt's baddbmm faulted on
+        # the stride-0 broadcast batch bias at this scale
+        h = torch.bmm(xg, self.w1.transpose(1, 2))
+        from ..ops import batched_bias_gelu
+        h = batched_bias_gelu(h, self.b1)
+        y = torch.bmm(h, self.w2.transpose(1, 2)) + self.b2.unsqueeze(1)
         valid = ar[None, :] < cnt[:, None]
         return y.reshape(-1, D)[valid.reshape(-1)]
 

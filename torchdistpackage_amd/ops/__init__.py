@@ -558,6 +558,40 @@ def cross_entropy_loss(logits: torch.Tensor, targets: torch.Tensor):
 # RoPE (rotate-half) and SwiGLU — fused Llama hot-path ops
 # ---------------------------------------------------------------------------
 
+class _BatchedBiasGeluFn(torch.autograd.Function):
+    """gelu(x + bias[e]) over (E, N, H) with per-expert bias — the MoE
+    batched-expert epilogue (hipBLASLt baddbmm faulted on a stride-0
+    broadcast batch bias; bmm + this fused kernel replaces it)."""
+
+    @staticmethod
+    def forward(ctx, x, bias):
+        ctx.save_for_backward(x, bias)
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            return ext("bgelu_b").bgelu_b_fwd(x.contiguous(),
+                                              bias.contiguous())
+        u = x.float() + bias.float().unsqueeze(1)
+        return _gelu_tanh(u).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, bias = ctx.saved_tensors
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            dx = ext("bgelu_b").bgelu_b_bwd(dy.contiguous(), x.contiguous(),
+                                            bias.contiguous())
+        else:
+            u = x.float() + bias.float().unsqueeze(1)
+            t = torch.tanh(_GELU_C * (u + 0.044715 * u.pow(3)))
+            du = 0.5 * (1 + t) + 0.5 * u * (1 - t * t) * _GELU_C * \
+                (1 + 3 * 0.044715 * u.pow(2))
+            dx = (dy.float() * du).to(x.dtype)
+        dbias = dx.float().sum(1).to(bias.dtype)
+        return dx, dbias
+
+
+def batched_bias_gelu(x, bias):
+    return _BatchedBiasGeluFn.apply(x, bias)
+
+
 class _RopeFn(torch.autograd.Function):
     """Rotate-half RoPE over (B, H, S, D) with fp32 (S, D/2) tables."""
 

@@ -44,6 +44,9 @@ torch::Tensor gemm_fprop(torch::Tensor x, torch::Tensor w,
 torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
                          long pos0, bool fwd);
 torch::Tensor swiglu_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor bgelu_b_fwd(torch::Tensor x, torch::Tensor bias);
+torch::Tensor bgelu_b_bwd(torch::Tensor dy, torch::Tensor x,
+                          torch::Tensor bias);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dy, torch::Tensor a,
                                       torch::Tensor b);
 torch::Tensor gemm_dgrad(torch::Tensor dy, torch::Tensor w, bool kswz);
@@ -75,6 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fprop", &gemm_fprop);
   m.def("rope_apply", &rope_apply);
   m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("bgelu_b_fwd", &bgelu_b_fwd);
+  m.def("bgelu_b_bwd", &bgelu_b_bwd);
   m.def("swiglu_bwd", &swiglu_bwd);
   m.def("gemm_dgrad", &gemm_dgrad);
   m.def("gemm_wgrad", &gemm_wgrad);

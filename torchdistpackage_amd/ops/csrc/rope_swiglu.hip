@@ -126,6 +126,61 @@ __global__ void swiglu_bwd_kernel(const unsigned short* __restrict__ dy,
   store8(db + g, fdb);
 }
 
+// per-expert bias + tanh-GELU over (E, N, H): bias index = (e, i % H).
+// Split out because hipBLASLt's baddbmm faulted on a stride-0 broadcast
+// batch bias at MoE bench scale (profiles/r02_notes.md) — the batched
+// expert path uses plain bmm + this fused epilogue instead.
+DEVINL float gelu_tanh_f(float u) {
+  float e = __builtin_exp2f((u + 0.044715f * u * u * u) *
+                            0.7978845608f * 2.885390082f);
+  float t = 1.f - 2.f / (e + 1.f);
+  return 0.5f * u * (1.f + t);
+}
+
+DEVINL float gelu_tanh_df(float u) {
+  float u2 = u * u;
+  float e = __builtin_exp2f((u + 0.044715f * u * u2) *
+                            0.7978845608f * 2.885390082f);
+  float t = 1.f - 2.f / (e + 1.f);
+  return 0.5f * (1.f + t) +
+         0.5f * u * (1.f - t * t) * 0.7978845608f *
+             (1.f + 3.f * 0.044715f * u2);
+}
+
+__global__ void bgelu_b_fwd_kernel(const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ bias,
+                                   unsigned short* __restrict__ y,
+                                   long n, long NH, int H) {
+  const long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (g >= n) return;
+  const long e = g / NH;
+  const int j = (int)(g % H);
+  float xv[8], bv[8], ov[8];
+  load8(x + g, xv);
+  load8(bias + e * H + j, bv);
+#pragma unroll
+  for (int k = 0; k < 8; ++k) ov[k] = gelu_tanh_f(xv[k] + bv[k]);
+  store8(y + g, ov);
+}
+
+__global__ void bgelu_b_bwd_kernel(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ x,
+                                   const unsigned short* __restrict__ bias,
+                                   unsigned short* __restrict__ dx,
+                                   long n, long NH, int H) {
+  const long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (g >= n) return;
+  const long e = g / NH;
+  const int j = (int)(g % H);
+  float dv[8], xv[8], bv[8], ov[8];
+  load8(dy + g, dv);
+  load8(x + g, xv);
+  load8(bias + e * H + j, bv);
+#pragma unroll
+  for (int k = 0; k < 8; ++k) ov[k] = dv[k] * gelu_tanh_df(xv[k] + bv[k]);
+  store8(dx + g, ov);
+}
+
 void check_bf16(const torch::Tensor& t, const char* n) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16 &&
               t.is_contiguous(), n, " must be contiguous bf16 CUDA");
@@ -191,4 +246,42 @@ std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dy, torch::Tensor a,
                      (unsigned short*)db.data_ptr(), n);
   HIP_CHECK_LAST();
   return {da, db};
+}
+
+// x (E, N, H) bf16 contiguous, bias (E, H) bf16; H % 8 == 0.
+torch::Tensor bgelu_b_fwd(torch::Tensor x, torch::Tensor bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+              x.dim() == 3 && x.is_contiguous() && bias.is_contiguous());
+  const int H = x.size(2);
+  TORCH_CHECK(H % 8 == 0 && bias.size(0) == x.size(0) &&
+              bias.size(1) == H);
+  auto y = torch::empty_like(x);
+  const long n = x.numel();
+  const long NH = (long)x.size(1) * H;
+  const long nb = (n / 8 + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(bgelu_b_fwd_kernel, dim3((unsigned)nb), dim3(256), 0,
+                     stream, (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)bias.data_ptr(),
+                     (unsigned short*)y.data_ptr(), n, NH, H);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor bgelu_b_bwd(torch::Tensor dy, torch::Tensor x,
+                          torch::Tensor bias) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  const int H = x.size(2);
+  auto dx = torch::empty_like(x);
+  const long n = x.numel();
+  const long NH = (long)x.size(1) * H;
+  const long nb = (n / 8 + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL(bgelu_b_bwd_kernel, dim3((unsigned)nb), dim3(256), 0,
+                     stream, (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)bias.data_ptr(),
+                     (unsigned short*)dx.data_ptr(), n, NH, H);
+  HIP_CHECK_LAST();
+  return dx;
 }
