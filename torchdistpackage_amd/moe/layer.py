@@ -89,6 +89,37 @@ class Expert(nn.Module):
         return fast_linear(h, self.fc2.weight, self.fc2.bias)
 
 
+class _ExpertGemms(torch.autograd.Function):
+    """y[e] = x[e] @ w[e]^T as a loop of 2-D mm into one preallocated
+    output.  torch.bmm (hipBLASLt batched-strided bf16) produced NaNs at
+    small shapes and memory faults at (8,2048+,2048)x(...,8192) on this
+    stack — plain 2-D GEMMs are the well-exercised path, and 2xE launches
+    per layer is still ~100x fewer than the per-expert-segment loop this
+    replaces."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        E, M, K = x.shape
+        N = w.shape[1]
+        y = torch.empty(E, M, N, dtype=x.dtype, device=x.device)
+        for e in range(E):
+            torch.mm(x[e], w[e].t(), out=y[e])
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        E, M, K = x.shape
+        dy = dy.contiguous()
+        dx = torch.empty_like(x)
+        dw = torch.empty_like(w)
+        for e in range(E):
+            torch.mm(dy[e], w[e], out=dx[e])
+            torch.mm(dy[e].t(), x[e], out=dw[e])
+        return dx, dw
+
+
 class BatchedExperts(nn.Module):
     """All local experts as stacked (E, ...) parameters, run as TWO
     baddbmm calls over every expert at once.
@@ -143,12 +174,10 @@ class BatchedExperts(nn.Module):
         # exactly this); padding rows are never read back, any row works
         idx = idx.clamp_(0, grouped.shape[0] - 1)
         xg = grouped.index_select(0, idx.reshape(-1)).view(E, maxn_pad, D)
-        # bmm + fused per-expert bias+GELU: hipBLASLt's baddbmm faulted on
-        # the stride-0 broadcast batch bias at this scale
-        h = torch.bmm(xg, self.w1.transpose(1, 2))
         from ..ops import batched_bias_gelu
+        h = _ExpertGemms.apply(xg, self.w1)
         h = batched_bias_gelu(h, self.b1)
-        y = torch.bmm(h, self.w2.transpose(1, 2)) + self.b2.unsqueeze(1)
+        y = _ExpertGemms.apply(h, self.w2) + self.b2.unsqueeze(1)
         valid = ar[None, :] < cnt[:, None]
         return y.reshape(-1, D)[valid.reshape(-1)]
 
