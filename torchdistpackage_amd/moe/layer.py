@@ -178,8 +178,17 @@ class BatchedExperts(nn.Module):
         h = _ExpertGemms.apply(xg, self.w1)
         h = batched_bias_gelu(h, self.b1)
         y = _ExpertGemms.apply(h, self.w2) + self.b2.unsqueeze(1)
-        valid = ar[None, :] < cnt[:, None]
-        return y.reshape(-1, D)[valid.reshape(-1)]
+        # sync-free valid-row extraction: boolean-mask indexing calls
+        # nonzero() -> a host sync per MoE layer (measured -17% end to
+        # end); the valid index list is constructible from cnt/offs with
+        # the host-known total
+        total = grouped.shape[0]
+        e_of = torch.repeat_interleave(
+            torch.arange(E, device=grouped.device), cnt, output_size=total)
+        pos = torch.arange(total, device=grouped.device) - offs.index_select(
+            0, e_of)
+        val_idx = e_of * maxn_pad + pos
+        return y.reshape(-1, D).index_select(0, val_idx)
 
 
 def _all_to_all_uneven(x: torch.Tensor, in_splits: List[int],
