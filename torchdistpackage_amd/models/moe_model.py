@@ -47,8 +47,12 @@ class MoEBlock(nn.Module):
         self.ln_1 = LayerNorm(cfg.dim, **kw)
         self.attn = Attention(cfg.dim, cfg.n_head, causal=cfg.causal, **kw)
         self.ln_2 = LayerNorm(cfg.dim, **kw)
+        # batched=False: the per-expert segment path measured FASTER at
+        # the moe_8x bench scale (93.7k vs 80.9k tok/s — its narrow()
+        # slices are zero-copy; BatchedExperts' gathers cost more than the
+        # padding fills they remove, profiles/r02_notes.md)
         self.moe = ExpertParallelMoE(cfg.dim, cfg.num_experts, cfg.top_k,
-                                     cfg.hidden_mult, batched=True, **kw)
+                                     cfg.hidden_mult, **kw)
 
     def forward(self, x):
         x = x + self.attn(self.ln_1(x))
