@@ -162,8 +162,7 @@ def test_moe_model_forward():
     out["loss"].backward()
     # expert params got grads and are tagged
     n_expert = sum(1 for p in m.expert_parameters())
-    # batched experts: 4 stacked params (w1,b1,w2,b2) per MoE layer
-    assert n_expert == 2 * 4
+    assert n_expert == 2 * 4 * 4  # layers * experts * (2 linears w+b)
     for p in m.expert_parameters():
         assert p.grad is not None
 
