@@ -15,7 +15,7 @@ mode here (SUM+div on gloo).
 
 from __future__ import annotations
 
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist

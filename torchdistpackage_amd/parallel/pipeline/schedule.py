@@ -15,7 +15,7 @@ per schedule (first micro-batch) and shapes are cached after.
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, List, Optional
 
 import torch
 

@@ -12,8 +12,6 @@ Layout: activations are sequence-first (S, B, D); SP shards dim 0.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

@@ -13,8 +13,6 @@ saving one full activation round-trip to HBM3E.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
