@@ -621,3 +621,18 @@ def test_vocab_parallel_ce_fused_vs_eager():
     ge = l_eager.grad
     gf = l_fused.grad.float()
     assert (gf - ge).abs().max() < 2e-3, (gf - ge).abs().max()
+
+
+def test_ema_update_bf16_large():
+    """bf16-param EMA at a size beyond ew_grid's 2048-block cap: every
+    element must update (a non-grid-strided version silently updated only
+    the first 524k of an 8B shard)."""
+    from torchdistpackage_amd.ops import ema_update_
+    n = 3_000_000   # > 2048 * 256
+    ema = torch.zeros(n, device=_dev())
+    p = torch.ones(n, device=_dev(), dtype=torch.bfloat16)
+    ema_update_(ema, p, 0.9)
+    expect = 0.1
+    assert abs(ema[0].item() - expect) < 1e-4
+    assert abs(ema[-1].item() - expect) < 1e-4, ema[-1].item()
+    assert abs(ema[600_000].item() - expect) < 1e-4

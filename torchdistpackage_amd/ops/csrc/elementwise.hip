@@ -270,8 +270,13 @@ namespace {
 __global__ void ema_bf16_kernel(float* __restrict__ ema,
                                 const unsigned short* __restrict__ p, long n,
                                 float decay) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) ema[i] = decay * ema[i] + (1.f - decay) * bf2f(p[i]);
+  // GRID-STRIDED: ew_grid caps launches at 2048 blocks — a plain
+  // one-element-per-thread body here updated only the first 524k elements
+  // of an 8B-param shard (caught by the large-n parity test)
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride)
+    ema[i] = decay * ema[i] + (1.f - decay) * bf2f(p[i]);
 }
 }  // namespace
 
