@@ -153,11 +153,23 @@ __global__ void adamw_kernel(float* __restrict__ p, const float* __restrict__ g,
 
 // -------- EMA: ema = d*ema + (1-d)*p (f32 ema; f32 OR bf16 p) ------------
 
+// vectorized 4-wide (the v10 adamw lesson: scalar dword accesses cap the
+// HBM-bound kernels well below the roofline); scalar tail for n % 4
 __global__ void ema_kernel(float* __restrict__ ema, const float* __restrict__ p,
-                           long n, float decay) {
+                           long n, float decay, int vec4) {
+  const long n4 = vec4 ? (n >> 2) : 0;
   long i0 = (long)blockIdx.x * BLOCK + threadIdx.x;
   long stride = (long)gridDim.x * BLOCK;
-  for (long i = i0; i < n; i += stride)
+  for (long i = i0; i < n4; i += stride) {
+    float4 e = *(const float4*)(ema + i * 4);
+    float4 v = *(const float4*)(p + i * 4);
+    e.x = decay * e.x + (1.f - decay) * v.x;
+    e.y = decay * e.y + (1.f - decay) * v.y;
+    e.z = decay * e.z + (1.f - decay) * v.z;
+    e.w = decay * e.w + (1.f - decay) * v.w;
+    *(float4*)(ema + i * 4) = e;
+  }
+  for (long i = n4 * 4 + i0; i < n; i += stride)
     ema[i] = decay * ema[i] + (1.f - decay) * p[i];
 }
 
@@ -269,13 +281,25 @@ void adamw_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 namespace {
 __global__ void ema_bf16_kernel(float* __restrict__ ema,
                                 const unsigned short* __restrict__ p, long n,
-                                float decay) {
+                                float decay, int vec4) {
   // GRID-STRIDED: ew_grid caps launches at 2048 blocks — a plain
   // one-element-per-thread body here updated only the first 524k elements
-  // of an 8B-param shard (caught by the large-n parity test)
+  // of an 8B-param shard (caught by the large-n parity test); 4-wide
+  // vectorized like the fp32 variant (vec4=0: unaligned narrow() views
+  // fall back to the scalar loop)
+  const long n4 = vec4 ? (n >> 2) : 0;
   const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
-  for (long i = i0; i < n; i += stride)
+  for (long i = i0; i < n4; i += stride) {
+    float4 e = *(const float4*)(ema + i * 4);
+    ushort4 v = *(const ushort4*)(p + i * 4);
+    e.x = decay * e.x + (1.f - decay) * bf2f(v.x);
+    e.y = decay * e.y + (1.f - decay) * bf2f(v.y);
+    e.z = decay * e.z + (1.f - decay) * bf2f(v.z);
+    e.w = decay * e.w + (1.f - decay) * bf2f(v.w);
+    *(float4*)(ema + i * 4) = e;
+  }
+  for (long i = n4 * 4 + i0; i < n; i += stride)
     ema[i] = decay * ema[i] + (1.f - decay) * bf2f(p[i]);
 }
 }  // namespace
@@ -284,17 +308,24 @@ void ema_update(torch::Tensor ema, torch::Tensor p, double decay) {
   TORCH_CHECK(ema.is_cuda() && ema.scalar_type() == torch::kFloat);
   long n = ema.numel();
   auto stream = at::cuda::getCurrentHIPStream();
+  // vector path needs 16B-aligned ema and 16B/8B-aligned p: EMA views are
+  // narrow() slices at arbitrary element offsets
+  const int vec4 = (((unsigned long long)ema.data_ptr() & 15) == 0 &&
+                    ((unsigned long long)p.data_ptr() &
+                     (p.scalar_type() == torch::kBFloat16 ? 7 : 15)) == 0)
+                   ? 1 : 0;
   if (p.scalar_type() == torch::kBFloat16) {
     // bf16 model params feed the fp32 EMA shard directly — the Python
     // lerp_(p.float()) fallback this replaces materialized an fp32 copy
     // of every param each step (r02 Llama profile: 14% eager elementwise)
-    hipLaunchKernelGGL(ema_bf16_kernel, dim3(ew_grid(n)), dim3(BLOCK), 0,
+    hipLaunchKernelGGL(ema_bf16_kernel, dim3(ew_grid(n, 4)), dim3(BLOCK), 0,
                        stream, ema.data_ptr<float>(),
-                       (const unsigned short*)p.data_ptr(), n, (float)decay);
+                       (const unsigned short*)p.data_ptr(), n, (float)decay,
+                       vec4);
   } else {
-    hipLaunchKernelGGL(ema_kernel, dim3(ew_grid(n)), dim3(BLOCK), 0, stream,
-                       ema.data_ptr<float>(), p.data_ptr<float>(), n,
-                       (float)decay);
+    hipLaunchKernelGGL(ema_kernel, dim3(ew_grid(n, 4)), dim3(BLOCK), 0,
+                       stream, ema.data_ptr<float>(), p.data_ptr<float>(), n,
+                       (float)decay, vec4);
   }
   HIP_CHECK_LAST();
 }
