@@ -76,6 +76,11 @@ class Bf16ZeroOptimizer:
         self.num_grad_acc_iter = max(1, num_grad_acc_iter)
         self._fires: Dict[int, int] = {}
 
+        # inner optimizers that read grads by pointer (FusedAdamW) accept a
+        # bf16 grad override on the fp32 masters — the cast-copy into
+        # mp.grad is skipped entirely (saves 2 full grad passes per step)
+        self._grad_override = getattr(optimizer, "supports_grad_override",
+                                      False)
         self.rank = dist.get_rank(self.group) if dist.is_initialized() else 0
         self.world = dist.get_world_size(self.group) if dist.is_initialized() else 1
 
@@ -208,6 +213,12 @@ class Bf16ZeroOptimizer:
     def _finish_reduction(self):
         if self._grad_world() == 1:
             # single-rank: no comm; master grads come straight from p.grad.
+            if self._grad_override:
+                for i in self._my_idx:
+                    p = self._params[i]
+                    if p.grad is not None:
+                        self._master_params[i]._tdpa_grad_override = p.grad
+                return
             # Batched (ONE foreach cast-copy): the per-param loop was ~300
             # bf16->fp32 kernel launches per step on Llama-8B (r02 profile)
             dsts, srcs = [], []
@@ -269,10 +280,14 @@ class Bf16ZeroOptimizer:
                 continue
             if self._owner[i] == self.rank:
                 mp = self._master_params[i]
-                if mp.grad is None:
-                    mp.grad = torch.empty_like(mp)
-                mp.grad.copy_(p.grad)
-            if self.stage2:
+                if self._grad_override:
+                    mp._tdpa_grad_override = p.grad
+                else:
+                    if mp.grad is None:
+                        mp.grad = torch.empty_like(mp)
+                    mp.grad.copy_(p.grad)
+            if self.stage2 and not (self._grad_override and
+                                    self._owner[i] == self.rank):
                 p.grad = None
         # copy (cast) owned grads into master grads; optionally free the rest.
         # Only params whose hook pushed THIS iteration: a never-fired param's
@@ -285,10 +300,14 @@ class Bf16ZeroOptimizer:
                 i = self._idx_of[id(p)]
                 if self._owner[i] == self.rank:
                     mp = self._master_params[i]
-                    if mp.grad is None:
-                        mp.grad = torch.empty_like(mp)
-                    dsts.append(mp.grad)
-                    srcs.append(v.view(p.shape))
+                    if self._grad_override:
+                        # the reduced bucket view IS the grad: no cast-copy
+                        mp._tdpa_grad_override = v.view(p.shape)
+                    else:
+                        if mp.grad is None:
+                            mp.grad = torch.empty_like(mp)
+                        dsts.append(mp.grad)
+                        srcs.append(v.view(p.shape))
                 if self.stage2:
                     p.grad = None
             b.clear_pushed()
@@ -297,20 +316,34 @@ class Bf16ZeroOptimizer:
 
     # ------------------------------------------------------------------
 
+    def _grad_of(self, i):
+        mp = self._master_params[i]
+        ov = getattr(mp, "_tdpa_grad_override", None)
+        return ov if ov is not None else mp.grad
+
+    def _clear_overrides(self):
+        if not self._grad_override:
+            return
+        for i in self._my_idx:
+            mp = self._master_params[i]
+            if getattr(mp, "_tdpa_grad_override", None) is not None:
+                mp._tdpa_grad_override = None
+
     @torch.no_grad()
     def step(self, closure=None):
         self._finish_reduction()
         if self.clip_grad > 0:
             self._clip_master_grads(self.clip_grad)
         self.optim.step()
+        self._clear_overrides()
         self._sync_params()
         return None
 
     @torch.no_grad()
     def _clip_master_grads(self, max_norm: float):
         from ..ops import l2norm_sq
-        grads = [self._master_params[i].grad for i in self._my_idx
-                 if self._master_params[i].grad is not None]
+        grads = [self._grad_of(i) for i in self._my_idx
+                 if self._grad_of(i) is not None]
         if grads:
             local_sq = torch.stack(
                 [l2norm_sq(g.reshape(-1)) for g in grads]).sum()
@@ -368,6 +401,7 @@ class Bf16ZeroOptimizer:
                 p.grad.zero_() if p.grad is not None else None)
         for i in self._my_idx:
             self._master_params[i].grad = None
+        self._clear_overrides()
 
     # pass-throughs ------------------------------------------------------
 
