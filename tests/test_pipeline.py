@@ -299,3 +299,10 @@ def _pp_tp_vocab_parallel_parity(rank, world_size):
 
 def test_pp2_tp2_vocab_parallel_parity():
     run_distributed(_pp_tp_vocab_parallel_parity, world_size=4)
+
+
+def test_1f1b_pp4():
+    """Deeper pipe: 3 mid stages exercise warmup depth 3 + fused steady
+    paths (the reference's pp>=3 bug class)."""
+    run_distributed(_pp_iteration, world_size=4,
+                    kwargs={"num_microbatches": 8})

@@ -18,7 +18,6 @@ from typing import List, Optional
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops import RMSNorm, flash_attention, rope_rotate_half, swiglu
 from ..ops.gemm import linear as fast_linear
