@@ -636,3 +636,35 @@ def test_ema_update_bf16_large():
     assert abs(ema[0].item() - expect) < 1e-4
     assert abs(ema[-1].item() - expect) < 1e-4, ema[-1].item()
     assert abs(ema[600_000].item() - expect) < 1e-4
+
+
+def test_gemm_shape_fuzz():
+    """Randomized eligible shapes through all three GEMM layouts vs the
+    fp32 oracle (tail slots, split-K, swizzle variants)."""
+    e = ops.ext("gemm")
+    g = torch.Generator().manual_seed(123)
+
+    def r(lo, hi, q):
+        return int(torch.randint(lo // q, hi // q + 1, (1,),
+                                 generator=g)) * q
+
+    for trial in range(8):
+        M, N = r(256, 1536, 256), r(256, 1536, 256)
+        K = r(32, 2048, 32)
+        x = (torch.randn(M, K, generator=g) * 0.5).bfloat16().to(_dev())
+        w = (torch.randn(N, K, generator=g) * 0.5).bfloat16().to(_dev())
+        ref = x.float() @ w.float().t()
+        _gemm_yardstick(e.gemm_fprop(x, w, None), ref, x @ w.t())
+        if K % 256 == 0:
+            dy = (torch.randn(M, N, generator=g) * 0.5).bfloat16().to(_dev())
+            w2 = (torch.randn(N, K, generator=g) * 0.5).bfloat16().to(_dev())
+            ref2 = dy.float() @ w2.float()
+            _gemm_yardstick(e.gemm_dgrad(dy, w2, True), ref2, dy @ w2)
+        T = r(256, 4096, 32)
+        dyt = (torch.randn(T, M, generator=g) * 0.5).bfloat16().to(_dev())
+        xt = (torch.randn(T, N, generator=g) * 0.5).bfloat16().to(_dev())
+        ref3 = dyt.float().t() @ xt.float()
+        for sk in (1, 2):
+            if T % (32 * sk) == 0:
+                _gemm_yardstick(e.gemm_wgrad(dyt, xt, sk, True), ref3,
+                                dyt.t() @ xt)
