@@ -187,6 +187,11 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         raise SystemExit("--zero shards optimizer state over the dp group, "
                          "which is wrong for expert-parallel params (each EP "
                          "rank owns different experts); use NaiveDdp+MoeDP")
+    if args.zero and pp > 1:
+        raise SystemExit("--zero is wired for the dp(/tp) layouts here; "
+                         "with pp>1 each stage holds different params and "
+                         "the bench's shard-group selection does not apply "
+                         "(compose Bf16ZeroOptimizer per stage directly)")
     if args.zero:
         # ZeRO owns grad reduction — model stays unwrapped (wrapping with
         # NaiveDdp too would leave BOTH hook sets firing).
