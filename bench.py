@@ -90,6 +90,11 @@ def main():
         tpc.setup_process_groups(
             [("data", dp), ("pipe", pp), ("tensor", tp)])
     use_pp = pp > 1
+    if args.zero and use_pp:
+        raise SystemExit("--zero is wired for the dp(/tp) layouts here; "
+                         "with pp>1 each stage holds different params and "
+                         "the bench's shard-group selection does not apply "
+                         "(compose Bf16ZeroOptimizer per stage directly)")
 
     fix_rand(tpc.get_dp_rank() if world > 1 else 0)
 
@@ -187,11 +192,7 @@ def run_dp_tp_bench(args, cfg, dev, dtype, dp, tp):
         raise SystemExit("--zero shards optimizer state over the dp group, "
                          "which is wrong for expert-parallel params (each EP "
                          "rank owns different experts); use NaiveDdp+MoeDP")
-    if args.zero and pp > 1:
-        raise SystemExit("--zero is wired for the dp(/tp) layouts here; "
-                         "with pp>1 each stage holds different params and "
-                         "the bench's shard-group selection does not apply "
-                         "(compose Bf16ZeroOptimizer per stage directly)")
+
     if args.zero:
         # ZeRO owns grad reduction — model stays unwrapped (wrapping with
         # NaiveDdp too would leave BOTH hook sets firing).
