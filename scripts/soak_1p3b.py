@@ -11,7 +11,9 @@ m = GPT2Model(gpt2_xl_1p3b(), device=dev, dtype=torch.bfloat16)
 opt = FusedAdamW(m.parameters(), lr=1e-4)
 x = torch.randint(0, 50304, (16, 1024), device=dev)
 losses, mems, times = [], [], []
-for it in range(100):
+import os
+N_STEPS = int(os.environ.get("SOAK_STEPS", "100"))
+for it in range(N_STEPS):
     t0 = time.perf_counter()
     loss = m(x, labels=x)["loss"]
     loss.backward()
