@@ -151,12 +151,6 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ q,
     __syncthreads();
   }
 
-  // T5 static form: the younger-dispatched wave half loses VALU
-  // arbitration to the older half at every segment start; one priority
-  // bump (wave-uniform scalar branch — a plain threadIdx test lowers to
-  // an exec-masked s_setprio that hits EVERY wave) removes the penalty
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
   int buf = 0;
   for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
     // ---- swapped QK^T: S^T[key][q] for 2 key-subtiles of 32
@@ -440,12 +434,6 @@ __global__ void attn_bwd_dq_v2_kernel(const unsigned short* __restrict__ q,
   stage_dq(0, 0);
   __syncthreads();
 
-  // T5 static form: the younger-dispatched wave half loses VALU
-  // arbitration to the older half at every segment start; one priority
-  // bump (wave-uniform scalar branch — a plain threadIdx test lowers to
-  // an exec-masked s_setprio that hits EVERY wave) removes the penalty
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
   int buf = 0;
   for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
 
@@ -724,12 +712,6 @@ __global__ void attn_bwd_dkdv_v2_kernel(
   if (qt_start + KV < S) load_regs(qt_start + KV);
   __syncthreads();
 
-  // T5 static form: the younger-dispatched wave half loses VALU
-  // arbitration to the older half at every segment start; one priority
-  // bump (wave-uniform scalar branch — a plain threadIdx test lowers to
-  // an exec-masked s_setprio that hits EVERY wave) removes the penalty
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
   int buf = 0;
   for (int qt0 = qt_start; qt0 < S; qt0 += KV) {
 
@@ -969,12 +951,6 @@ __global__ void attn_bwd_dq_lite_kernel(
   if (KV < kv_end) load_regs(KV);
   __syncthreads();
 
-  // T5 static form: the younger-dispatched wave half loses VALU
-  // arbitration to the older half at every segment start; one priority
-  // bump (wave-uniform scalar branch — a plain threadIdx test lowers to
-  // an exec-masked s_setprio that hits EVERY wave) removes the penalty
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
   int buf = 0;
   for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
     bf16x8_v db[2][2];
@@ -1109,12 +1085,6 @@ __global__ void attn_bwd_dq_lite_tr_kernel(
   if (KV < kv_end) load_regs(KV);
   __syncthreads();
 
-  // T5 static form: the younger-dispatched wave half loses VALU
-  // arbitration to the older half at every segment start; one priority
-  // bump (wave-uniform scalar branch — a plain threadIdx test lowers to
-  // an exec-masked s_setprio that hits EVERY wave) removes the penalty
-  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
-    __builtin_amdgcn_s_setprio(1);
   int buf = 0;
   for (int kt0 = 0; kt0 < kv_end; kt0 += KV) {
     const unsigned abase = tr_base + (unsigned)(buf * 2 * 8 * KV * 16);
