@@ -41,3 +41,16 @@ def test_bench_json_contract_dp2():
 def test_bench_json_contract_8rank():
     p = _run_bench(8, 29602, ["--model", "tiny", "--batch", "2"])
     assert p["config"]["parallelism"] == "dp2_pp2_tp2"
+
+
+@pytest.mark.slow
+def test_bench_json_contract_zero_tp():
+    """ZeRO + TP composition through the real bench flow (the r01-advisor
+    bug class: SP grads must survive ZeRO's bucketing)."""
+    os.environ["TDPA_PARALLEL"] = "2,1,2"
+    try:
+        p = _run_bench(4, 29603, ["--model", "tiny", "--batch", "4",
+                                  "--zero"])
+    finally:
+        os.environ.pop("TDPA_PARALLEL", None)
+    assert p["config"]["parallelism"] == "dp2_pp1_tp2"
