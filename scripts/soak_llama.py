@@ -15,8 +15,9 @@ inner = FusedAdamW(m.parameters(), lr=1e-4)
 opt = Bf16ZeroOptimizer(inner, stage2=True)
 ema = ShardedEMA(m, decay=0.999)
 x = torch.randint(0, 128256, (4, 1024), device=dev)
+N_STEPS = int(os.environ.get("SOAK_STEPS", "100"))
 losses, mems = [], []
-for it in range(100):
+for it in range(N_STEPS):
     loss = m(x, labels=x)["loss"]
     loss.backward()
     opt.step(); opt.zero_grad(); ema.update()

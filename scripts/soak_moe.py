@@ -11,8 +11,9 @@ torch.manual_seed(0)
 m = MoEModel(mixtral_style_8x(), device=dev, dtype=torch.bfloat16)
 opt = FusedAdamW(m.parameters(), lr=1e-4)
 x = torch.randint(0, 50304, (16, 1024), device=dev)
+N_STEPS = int(os.environ.get("SOAK_STEPS", "100"))
 losses, mems = [], []
-for it in range(100):
+for it in range(N_STEPS):
     loss = m(x, labels=x)["loss"]
     loss.backward()
     opt.step(); opt.zero_grad()
