@@ -2,7 +2,10 @@
 expert-parallel all-to-all + MoE-DP replicas over xGMI).
 
 GPT-2 backbone with every block's dense MLP replaced by an
-ExpertParallelMoE layer; attention/norms identical to models.gpt2.
+ExpertParallelMoE layer; attention/norms identical to models.gpt2.  The
+reference's MoE recipe swaps ``block.mlp`` for DeepSpeed's MoE layer
+(/root/reference/explore/moe/ds_fmoe_main.py:22-25) — here the dispatch
+layer is in-package (moe/layer.py) and the swap is a config flag.
 """
 
 from __future__ import annotations
