@@ -49,3 +49,6 @@ def main():
 
 if __name__ == "__main__":
     main()
+    import torch.distributed as _d
+    if _d.is_initialized():
+        _d.destroy_process_group()
