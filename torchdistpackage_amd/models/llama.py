@@ -128,6 +128,35 @@ class LlamaAttention(nn.Module):
             o = o.permute(2, 0, 1, 3).reshape(S, B, self.nh_local * self.hd)
         return self.wo(o)
 
+    @torch.no_grad()
+    def decode_step(self, x: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, pos0: int) -> torch.Tensor:
+        """KV-cache inference step (tp=1 only; see inference/generate.py).
+        RoPE is applied at the absolute positions ``pos0..pos0+S_new``; the
+        cache holds rotated K, so cached entries are reused verbatim.  K/V
+        heads are expanded to Q heads for eager SDPA (decode is GEMV-shaped,
+        memory-bound — the flash kernel is the training-shape path)."""
+        from ..parallel.tensor.attn import _cached_sdpa
+        S, B, _ = x.shape
+        q = fast_linear(x, self.wq.weight)
+        k = fast_linear(x, self.wk.weight)
+        v = fast_linear(x, self.wv.weight)
+
+        def view4(t, nh):
+            return t.reshape(S, B, nh, self.hd).permute(1, 2, 0, 3)
+
+        q = self.rope(view4(q, self.nh_local), pos0)
+        k = self.rope(view4(k, self.nkv_local), pos0)
+        v = view4(v, self.nkv_local)
+        k_cache[:, :, pos0:pos0 + S] = k
+        v_cache[:, :, pos0:pos0 + S] = v
+        rep = self.nh_local // self.nkv_local
+        o = _cached_sdpa(q, k_cache.repeat_interleave(rep, dim=1),
+                         v_cache.repeat_interleave(rep, dim=1), pos0,
+                         self.causal)
+        o = o.permute(2, 0, 1, 3).reshape(S, B, self.nh_local * self.hd)
+        return self.wo(o)
+
 
 class LlamaMlp(nn.Module):
     """SwiGLU: w2(silu(w1 x) * w3 x), TP col/col/row."""
@@ -178,6 +207,12 @@ class LlamaBlock(nn.Module):
         if self.sequence_parallel:
             set_sequence_parallel_attr(x)
         return x
+
+    @torch.no_grad()
+    def decode_step(self, x, k_cache, v_cache, pos0: int):
+        x = x + self.attn.decode_step(self.attn_norm(x), k_cache, v_cache,
+                                      pos0)
+        return x + self.mlp(self.mlp_norm(x))
 
 
 class LlamaEmbedding(nn.Module):

@@ -22,6 +22,25 @@ from .tp_utils import (ColParallelLinear, RowParallelLinear, TpLinear,
                        get_tp_size, is_sequence_parallel)
 
 
+def _cached_sdpa(q: torch.Tensor, k_cache: torch.Tensor,
+                 v_cache: torch.Tensor, pos0: int,
+                 causal: bool) -> torch.Tensor:
+    """Attention of q (B, H, S_new, hd) against cache positions
+    [0, pos0+S_new), with the causal mask aligned so new position i sees
+    cached positions <= pos0 + i (torch's ``is_causal`` is top-left
+    aligned, wrong for a chunk appended mid-sequence)."""
+    S = q.shape[2]
+    ka = k_cache[:, :, :pos0 + S]
+    va = v_cache[:, :, :pos0 + S]
+    if not causal or S == 1:
+        return F.scaled_dot_product_attention(q, ka, va)
+    if pos0 == 0:
+        return F.scaled_dot_product_attention(q, ka, va, is_causal=True)
+    mask = torch.arange(pos0 + S, device=q.device)[None, :] <= \
+        (pos0 + torch.arange(S, device=q.device))[:, None]
+    return F.scaled_dot_product_attention(q, ka, va, attn_mask=mask)
+
+
 def _sdpa(x_qkv: torch.Tensor, n_head: int, causal: bool) -> torch.Tensor:
     """(S, B, 3*Hl*hd) fused qkv -> (S, B, Hl*hd) attention output.
 
@@ -80,6 +99,34 @@ class TpAttention(nn.Module):
         # qkv layout per rank: [q_local | k_local | v_local] thanks to the
         # interleaved loader (init_qkv_weight_from_full) / native init
         o = _sdpa(qkv, self.n_head_local, self.causal)
+        return self.proj(o)
+
+    @torch.no_grad()
+    def decode_step(self, x: torch.Tensor, k_cache: torch.Tensor,
+                    v_cache: torch.Tensor, pos0: int) -> torch.Tensor:
+        """KV-cache inference step (tp=1 only; see inference/generate.py).
+
+        x (S_new, B, D) — the new tokens' hidden states; writes their K/V
+        into ``k_cache``/``v_cache`` (B, H, max_seq, hd) at ``pos0`` and
+        attends q against positions [0, pos0+S_new).  Decode is
+        memory-bound GEMV-shaped work, so eager SDPA over the cache is the
+        right tool (the flash kernel is the training-shape path).
+        """
+        from ...ops.gemm import linear as fast_linear
+        S, B, D = x.shape
+        hl = self.n_head_local
+        hd = D // self.n_head
+        qkv = fast_linear(x, self.qkv.weight, self.qkv.bias)
+        q, k, v = qkv.split(hl * hd, dim=-1)
+
+        def v4(t):
+            return t.reshape(S, B, hl, hd).permute(1, 2, 0, 3)
+
+        q, k, v = v4(q), v4(k), v4(v)
+        k_cache[:, :, pos0:pos0 + S] = k
+        v_cache[:, :, pos0:pos0 + S] = v
+        o = _cached_sdpa(q, k_cache, v_cache, pos0, self.causal)
+        o = o.permute(2, 0, 1, 3).reshape(S, B, hl * hd)
         return self.proj(o)
 
     @torch.no_grad()

@@ -92,6 +92,14 @@ class ParallelBlock(nn.Module):
         return x
 
     @torch.no_grad()
+    def decode_step(self, x, k_cache, v_cache, pos0: int):
+        """KV-cache inference step (tp=1 only; see inference/generate.py):
+        the training forward's SP plumbing is inert at tp=1, so LN/MLP are
+        reused as-is and only attention takes the cache."""
+        x = x + self.attn.decode_step(self.ln_1(x), k_cache, v_cache, pos0)
+        return x + self.mlp(self.ln_2(x))
+
+    @torch.no_grad()
     def init_from_full(self, full: Block):
         self.ln_1.load_state_dict(full.ln_1.state_dict())
         self.ln_2.load_state_dict(full.ln_2.state_dict())
