@@ -26,7 +26,7 @@ def test_decode_logits_match_training_path():
                            torch.device("cuda"), torch.bfloat16)
     # prefill 8 tokens, then single-token steps, teacher-forced
     pos0, chunk = 0, toks[:, :8]
-    while pos0 + chunk.shape[1] <= T:
+    while chunk.shape[1] > 0:
         logits = _gpt2_decode_forward(m, chunk, caches, pos0).float()
         pos = pos0 + chunk.shape[1] - 1
         diff = (logits - full[:, pos]).abs().max().item()
