@@ -54,6 +54,21 @@ with torch.no_grad():
 
 print(f"prefill {B}x{PROMPT}: {t_prefill*1e3:.1f} ms "
       f"({B*PROMPT/t_prefill:.0f} tok/s)")
-print(f"decode: {dt/n_timed*1e3:.2f} ms/step, "
+print(f"eager decode: {dt/n_timed*1e3:.2f} ms/step, "
+      f"{B*n_timed/dt:.0f} tok/s (batch {B})")
+
+from torchdistpackage_amd.inference.generate import GraphedGPT2Decoder
+
+dec = GraphedGPT2Decoder(m, batch=B, max_seq=PROMPT + NEW)
+dec.prefill(idx)
+for _ in range(8):
+    dec.step()
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+for _ in range(n_timed):
+    dec.step()
+torch.cuda.synchronize()
+dt = time.perf_counter() - t0
+print(f"graphed decode: {dt/n_timed*1e3:.2f} ms/step, "
       f"{B*n_timed/dt:.0f} tok/s (batch {B})")
 print("DECODE BENCH OK")
