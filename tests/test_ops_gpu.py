@@ -668,3 +668,24 @@ def test_gemm_shape_fuzz():
             if T % (32 * sk) == 0:
                 _gemm_yardstick(e.gemm_wgrad(dyt, xt, sk, True), ref3,
                                 dyt.t() @ xt)
+
+
+@pytest.mark.parametrize("M,N,K", [(1, 2048, 2048), (2, 512, 1024),
+                                   (7, 6144, 2048), (16, 2048, 8192),
+                                   (16, 50304, 2048), (32, 1000, 4096),
+                                   (5, 2048, 2056)])
+def test_gemv_decode(M, N, K):
+    """Skinny-M streaming GEMV (decode path) vs fp32 reference."""
+    from torchdistpackage_amd.ops import ext
+    g = torch.Generator().manual_seed(M * 31 + N)
+    x = (torch.randn(M, K, generator=g) * 0.5).bfloat16().to(_dev())
+    w = (torch.randn(N, K, generator=g) * 0.5).bfloat16().to(_dev())
+    b = (torch.randn(N, generator=g) * 0.5).bfloat16().to(_dev())
+    ref = x.float() @ w.float().t() + b.float()
+    lib = torch.nn.functional.linear(x, w, b).float()
+    got = ext("gemv").gemv_bf16(x, w, b).float()
+    assert (got - ref).abs().max() <= \
+        (lib - ref).abs().max().clamp_min(0.1) * 2.0
+    got_nb = ext("gemv").gemv_bf16(x, w, None).float()
+    ref_nb = x.float() @ w.float().t()
+    assert (got_nb - ref_nb).abs().max() < 0.35

@@ -26,6 +26,7 @@ SOURCES = [
     "probe.hip",
     "gemm.hip",
     "rope_swiglu.hip",
+    "gemv.hip",
 ]
 
 

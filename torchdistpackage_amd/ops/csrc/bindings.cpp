@@ -56,6 +56,8 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
                      torch::Tensor lse, torch::Tensor grad_out);
 std::vector<torch::Tensor> ce_partial_fwd(torch::Tensor logits,
                                           torch::Tensor targets);
+torch::Tensor gemv_bf16(torch::Tensor x, torch::Tensor W,
+                        c10::optional<torch::Tensor> bias);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -85,4 +87,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_wgrad", &gemm_wgrad);
   m.def("ce_bwd", &ce_bwd);
   m.def("ce_partial_fwd", &ce_partial_fwd);
+  m.def("gemv_bf16", &gemv_bf16);
 }

@@ -37,6 +37,7 @@ import torch.nn.functional as F
 from . import ext
 
 _MODE = os.environ.get("TDPA_GEMM", "1")
+_GEMV = os.environ.get("TDPA_GEMV", "1") == "1"
 # wgrad K=16384-class shapes keep fp32 partials in a slab; cap its size
 _MAX_SLAB_BYTES = 2 << 30
 
@@ -121,6 +122,16 @@ def linear(x: torch.Tensor, weight: torch.Tensor,
         M = x.numel() // shape[-1]
         K = shape[-1]
         N = weight.shape[0]
+        # decode path (M <= 32 tokens, inference only): the in-tree
+        # streaming GEMV (csrc/gemv.hip); TDPA_GEMV=0 reverts to hipBLASLt
+        if (_GEMV and M <= 32 and K % 8 == 0
+                and not torch.is_grad_enabled()
+                and weight.is_contiguous()):
+            x2d = x.reshape(M, K)
+            if not x2d.is_contiguous():
+                x2d = x2d.contiguous()
+            out = ext("gemv").gemv_bf16(x2d, weight, bias)
+            return out.reshape(*shape[:-1], N)
         if _eligible(M, N, K) and (
                 _MODE == "all"
                 or (K % 256 == 0 and (_use_mine("wgrad", N, K, M)
