@@ -1,12 +1,12 @@
 // Skinny-M GEMV for decode: y[M,N] = x[M,K] @ W[N,K]^T (+bias), bf16.
 //
 // Inference decode (inference/generate.py) is bound by streaming the
-// weight matrices once per token; hipBLASLt's skinny-M kernels measured
-// ~1 TB/s effective on these shapes (2.54 ms/step at 1.3B).  This kernel
-// is a pure streaming design: one WAVE per output row n reads W[n,:] with
-// coalesced 16-byte loads at full HBM rate; the (tiny, hot) x rows are
-// staged through LDS once per workgroup and reused by all 4 waves; fp32
-// accumulate, one cross-lane reduction per output.
+// weight matrices once per token.  Pure streaming design: one WAVE
+// computes TWO output rows n (so every LDS read of x is amortized over
+// 2x the weight bytes — the v1 one-row variant was LDS/VALU-bound, not
+// HBM-bound), reading W with coalesced 16-byte loads; the (tiny, hot) x
+// rows are staged through LDS once per workgroup; products accumulate in
+// fp32 via v_dot2_f32_bf16 (2 bf16 MACs per VALU op, no unpack).
 //
 // M <= 32 (decode batch), K % 8 == 0, row-major contiguous x/W/y.
 #include <torch/extension.h>
@@ -14,7 +14,9 @@
 
 #include "common.h"
 
-#define GEMV_BLOCK 256  // 4 waves; wave w handles n = blockIdx.x*4 + w
+#define GEMV_BLOCK 256  // 4 waves; wave w handles rows 2*(blockIdx.x*4+w)+{0,1}
+
+typedef short sv2 __attribute__((ext_vector_type(2)));
 
 template <int MT>
 __global__ __launch_bounds__(GEMV_BLOCK) void gemv_bf16_kernel(
@@ -25,10 +27,10 @@ __global__ __launch_bounds__(GEMV_BLOCK) void gemv_bf16_kernel(
   __shared__ ushort xs[MT * KT];
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const long n = (long)blockIdx.x * 4 + wave;
-  float acc[MT];
+  const long n0 = ((long)blockIdx.x * 4 + wave) * 2;
+  float acc0[MT], acc1[MT];
 #pragma unroll
-  for (int m = 0; m < MT; ++m) acc[m] = 0.f;
+  for (int m = 0; m < MT; ++m) acc0[m] = acc1[m] = 0.f;
 
   for (long k0 = 0; k0 < K; k0 += KT) {
     const int kt = (int)min((long)KT, K - k0);
@@ -38,35 +40,50 @@ __global__ __launch_bounds__(GEMV_BLOCK) void gemv_bf16_kernel(
       *(uint4*)&xs[m * KT + c] = *(const uint4*)&x[(long)m * K + k0 + c];
     }
     __syncthreads();
-    if (n < N) {
-      const ushort* wrow = W + n * K + k0;
+    if (n0 < N) {
+      const ushort* w0 = W + n0 * K + k0;
+      const bool two = (n0 + 1) < N;
+      const ushort* w1 = two ? w0 + K : w0;
       for (int kk = lane * 8; kk < kt; kk += WAVE * 8) {
-        const uint4 wv = *(const uint4*)&wrow[kk];
-        const ushort* wu = (const ushort*)&wv;
-        float wf[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) wf[j] = bf2f(wu[j]);
+        const uint4 wa = *(const uint4*)&w0[kk];
+        const uint4 wb = *(const uint4*)&w1[kk];
+        const sv2* wpa = (const sv2*)&wa;
+        const sv2* wpb = (const sv2*)&wb;
 #pragma unroll
         for (int m = 0; m < MT; ++m) {
           if (m < M) {
             const uint4 xv = *(const uint4*)&xs[m * KT + kk];
-            const ushort* xu = (const ushort*)&xv;
+            const sv2* xp = (const sv2*)&xv;
 #pragma unroll
-            for (int j = 0; j < 8; ++j) acc[m] += wf[j] * bf2f(xu[j]);
+            for (int j = 0; j < 4; ++j) {
+              acc0[m] = __builtin_amdgcn_fdot2_f32_bf16(wpa[j], xp[j],
+                                                        acc0[m], false);
+              acc1[m] = __builtin_amdgcn_fdot2_f32_bf16(wpb[j], xp[j],
+                                                        acc1[m], false);
+            }
           }
         }
       }
     }
     __syncthreads();
   }
-  if (n >= N) return;
+  if (n0 >= N) return;
 #pragma unroll
-  for (int m = 0; m < MT; ++m) acc[m] = wave_sum(acc[m]);
+  for (int m = 0; m < MT; ++m) {
+    acc0[m] = wave_sum(acc0[m]);
+    acc1[m] = wave_sum(acc1[m]);
+  }
   if (lane == 0) {
-    const float b = has_bias ? bf2f(bias[n]) : 0.f;
+    const float b0 = has_bias ? bf2f(bias[n0]) : 0.f;
 #pragma unroll
     for (int m = 0; m < MT; ++m)
-      if (m < M) y[(long)m * N + n] = f2bf(acc[m] + b);
+      if (m < M) y[(long)m * N + n0] = f2bf(acc0[m] + b0);
+    if (n0 + 1 < N) {
+      const float b1 = has_bias ? bf2f(bias[n0 + 1]) : 0.f;
+#pragma unroll
+      for (int m = 0; m < MT; ++m)
+        if (m < M) y[(long)m * N + n0 + 1] = f2bf(acc1[m] + b1);
+    }
   }
 }
 
@@ -90,7 +107,7 @@ torch::Tensor gemv_bf16(torch::Tensor x, torch::Tensor W,
     has_bias = 1;
   }
   auto y = torch::empty({M, N}, x.options());
-  const dim3 grid((unsigned)((N + 3) / 4)), block(GEMV_BLOCK);
+  const dim3 grid((unsigned)((N + 7) / 8)), block(GEMV_BLOCK);
   auto stream = at::cuda::getCurrentHIPStream();
   const ushort* xp = (const ushort*)x.data_ptr();
   const ushort* wp = (const ushort*)W.data_ptr();
