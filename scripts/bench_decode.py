@@ -18,6 +18,7 @@ from torchdistpackage_amd.models.gpt2 import GPT2Model, gpt2_xl_1p3b
 B = int(os.environ.get("DEC_BATCH", "16"))
 PROMPT = int(os.environ.get("DEC_PROMPT", "128"))
 NEW = int(os.environ.get("DEC_NEW", "128"))
+B1 = os.environ.get("DEC_B1", "1") == "1"   # also run the batch-1 latency case
 
 dev = torch.device("cuda")
 torch.manual_seed(0)
@@ -71,4 +72,19 @@ torch.cuda.synchronize()
 dt = time.perf_counter() - t0
 print(f"graphed decode: {dt/n_timed*1e3:.2f} ms/step, "
       f"{B*n_timed/dt:.0f} tok/s (batch {B})")
+
+if B1:
+    idx1 = idx[:1]
+    dec1 = GraphedGPT2Decoder(m, batch=1, max_seq=PROMPT + NEW)
+    dec1.prefill(idx1)
+    for _ in range(8):
+        dec1.step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(n_timed):
+        dec1.step()
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"graphed decode B=1: {dt/n_timed*1e3:.2f} ms/step, "
+          f"{n_timed/dt:.0f} tok/s (single stream)")
 print("DECODE BENCH OK")

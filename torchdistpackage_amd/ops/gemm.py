@@ -64,6 +64,17 @@ def pick_splitk(M: int, N: int, K: int) -> int:
     return best
 
 
+def _use_gemv(M: int, N: int, K: int) -> bool:
+    """Measured crossover (scripts/kbench_gemv.py, profiles/r02_gemv.log):
+    the streaming GEMV wins while its ~linear-in-M time stays under
+    hipBLASLt's flat ~18 us launch floor."""
+    if M <= 4 and N <= 16384:
+        return True
+    if M <= 8 and K <= 4096 and N <= 16384:
+        return True
+    return M <= 16 and N <= 2048 and K <= 2048
+
+
 def _use_mine(kind: str, M: int, N: int, K: int) -> bool:
     if _MODE == "all":
         return True
@@ -122,9 +133,13 @@ def linear(x: torch.Tensor, weight: torch.Tensor,
         M = x.numel() // shape[-1]
         K = shape[-1]
         N = weight.shape[0]
-        # decode path (M <= 32 tokens, inference only): the in-tree
-        # streaming GEMV (csrc/gemv.hip); TDPA_GEMV=0 reverts to hipBLASLt
-        if (_GEMV and M <= 32 and K % 8 == 0
+        # decode path (small-M tokens, inference only): the in-tree
+        # streaming GEMV (csrc/gemv.hip) reads weights at 5-7.6 TB/s where
+        # hipBLASLt's skinny-M kernels have a flat ~18 us floor; the kernel's
+        # cost grows ~linearly in M (it turns LDS/VALU-bound), so dispatch
+        # follows the measured crossover table (profiles/r02_gemv.log).
+        # TDPA_GEMV=0 reverts to hipBLASLt everywhere.
+        if (_GEMV and K % 8 == 0 and _use_gemv(M, N, K)
                 and not torch.is_grad_enabled()
                 and weight.is_contiguous()):
             x2d = x.reshape(M, K)
