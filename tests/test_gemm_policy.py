@@ -86,3 +86,15 @@ def test_linear_cpu_fallback_exact():
     b = torch.randn(768)
     assert torch.equal(g.linear(x, w, b),
                        torch.nn.functional.linear(x, w, b))
+
+
+def test_gemv_crossover_policy():
+    g = _reload("1")
+    # batch-1 latency decode: every non-head projection routes to the GEMV
+    assert g._use_gemv(1, 6144, 2048)
+    assert g._use_gemv(1, 2048, 8192)
+    assert g._use_gemv(4, 8192, 2048)
+    assert g._use_gemv(8, 6144, 2048)     # K<=4096 family holds to M=8
+    assert not g._use_gemv(8, 2048, 8192)  # deep-K falls off after M=4
+    assert not g._use_gemv(16, 2048, 2048)  # in-graph AB: library wins
+    assert not g._use_gemv(1, 50304, 2048)  # vocab head: library

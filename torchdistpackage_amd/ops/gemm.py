@@ -67,12 +67,13 @@ def pick_splitk(M: int, N: int, K: int) -> int:
 def _use_gemv(M: int, N: int, K: int) -> bool:
     """Measured crossover (scripts/kbench_gemv.py, profiles/r02_gemv.log):
     the streaming GEMV wins while its ~linear-in-M time stays under
-    hipBLASLt's flat ~18 us launch floor."""
+    hipBLASLt's flat ~18 us launch floor.  The isolated microbench also
+    shows a proj-shape win at M=16, but inside the captured decode graph
+    that substitution measured -5% end to end (r02_decode AB) — the
+    policy follows the end-to-end number and stops at M=8."""
     if M <= 4 and N <= 16384:
         return True
-    if M <= 8 and K <= 4096 and N <= 16384:
-        return True
-    return M <= 16 and N <= 2048 and K <= 2048
+    return M <= 8 and K <= 4096 and N <= 16384
 
 
 def _use_mine(kind: str, M: int, N: int, K: int) -> bool:
