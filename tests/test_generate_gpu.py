@@ -64,3 +64,34 @@ def test_generate_runs_on_gpu():
     assert out.shape == (4, 48)
     out2 = generate(m, idx, 16, greedy=False, temperature=0.9, top_k=50)
     assert out2.shape == (4, 32)
+
+
+def test_rope_device_pos_matches_host_pos():
+    from torchdistpackage_amd.ops import rope_rotate_half
+    torch.manual_seed(0)
+    x = torch.randn(2, 4, 1, 64, device="cuda", dtype=torch.bfloat16)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, 64, 2).float() / 64))
+    freqs = torch.outer(torch.arange(128).float(), inv).cuda()
+    cos, sin = freqs.cos(), freqs.sin()
+    with torch.no_grad():
+        want = rope_rotate_half(x, cos, sin, 37)
+        got = rope_rotate_half(x, cos, sin,
+                               torch.tensor([37], device="cuda"))
+    assert torch.equal(got, want)
+
+
+def test_graphed_llama_decoder_vs_eager():
+    from torchdistpackage_amd.inference.generate import GraphedLlamaDecoder
+    from torchdistpackage_amd.models.llama import LlamaConfig, LlamaModel
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=2048, n_layer=3, n_head=8, n_kv_head=2,
+                      dim=512, ffn_dim=1024, max_seq=64)
+    m = LlamaModel(cfg, device="cuda", dtype=torch.bfloat16).eval()
+    idx = torch.randint(0, cfg.vocab_size, (2, 6), device="cuda")
+    dec = GraphedLlamaDecoder(m, batch=2, max_seq=40)
+    out_g = dec.generate(idx, 20)
+    out_e = generate(m, idx, 20)
+    assert out_g.shape == out_e.shape == (2, 26)
+    assert torch.equal(out_g[:, :12], out_e[:, :12])
+    agree = (out_g == out_e).float().mean().item()
+    assert agree >= 0.8, agree

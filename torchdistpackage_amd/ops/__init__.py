@@ -629,9 +629,20 @@ class _RopeFn(torch.autograd.Function):
 
 
 def rope_rotate_half(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
-                     pos0: int = 0) -> torch.Tensor:
+                     pos0=0) -> torch.Tensor:
     """Fused rotate-half rotary embedding (one read+write pass; the eager
-    form is 2 cats + 4 muls per call, models/llama.py r01)."""
+    form is 2 cats + 4 muls per call, models/llama.py r01).
+
+    ``pos0`` may be a 1-element long DEVICE tensor (inference only): the
+    kernel reads the position at run time, which keeps the op
+    hipGraph-capturable (inference/generate.py's graphed Llama decoder
+    advances the tensor in place between replays)."""
+    if torch.is_tensor(pos0):
+        assert not torch.is_grad_enabled(), \
+            "tensor pos0 is the no-grad decode path"
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            return ext("rope").rope_apply_pos(x, cos, sin, pos0, True)
+        return _RopeFn.apply(x, cos, sin, int(pos0.item()))
     return _RopeFn.apply(x, cos, sin, pos0)
 
 

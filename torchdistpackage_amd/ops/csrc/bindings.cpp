@@ -58,6 +58,8 @@ std::vector<torch::Tensor> ce_partial_fwd(torch::Tensor logits,
                                           torch::Tensor targets);
 torch::Tensor gemv_bf16(torch::Tensor x, torch::Tensor W,
                         c10::optional<torch::Tensor> bias);
+torch::Tensor rope_apply_pos(torch::Tensor x, torch::Tensor cs,
+                             torch::Tensor sn, torch::Tensor pos, bool fwd);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -88,4 +90,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("ce_partial_fwd", &ce_partial_fwd);
   m.def("gemv_bf16", &gemv_bf16);
+  m.def("rope_apply_pos", &rope_apply_pos);
 }

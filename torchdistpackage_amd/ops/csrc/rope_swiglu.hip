@@ -45,15 +45,19 @@ __global__ void rope_kernel(const unsigned short* __restrict__ x,
                             const float* __restrict__ cs,   // (S, D/2) cos
                             const float* __restrict__ sn,   // (S, D/2) sin
                             long rows, int S, int D, int pos0,
-                            long H, long xsb, long xsh, long xss) {
+                            long H, long xsb, long xsh, long xss,
+                            const long* __restrict__ pos0p) {
   const int half = D >> 1;
   const int chunks = half >> 3;                 // 8 elems per thread-chunk
   const long total = rows * chunks;
+  // pos0p: optional DEVICE position (hipGraph-capturable decode — the
+  // graph replays with the position tensor advanced in place)
+  const int p0 = pos0p ? (int)*pos0p : pos0;
   for (long g = (long)blockIdx.x * blockDim.x + threadIdx.x; g < total;
        g += (long)gridDim.x * blockDim.x) {
     const long row = g / chunks;
     const int c = (int)(g - row * chunks) * 8;
-    const int s = (int)(row % S) + pos0;
+    const int s = (int)(row % S) + p0;
     const long bh = row / S;
     const long xbase = (bh / H) * xsb + (bh - (bh / H) * H) * xsh +
                        (row % S) * xss + c;
@@ -210,7 +214,39 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cs, torch::Tensor sn,
                      (const unsigned short*)x.data_ptr(),
                      (unsigned short*)y.data_ptr(), cs.data_ptr<float>(),
                      sn.data_ptr<float>(), rows, S, D, (int)pos0,
-                     x.size(1), x.stride(0), x.stride(1), x.stride(2));
+                     x.size(1), x.stride(0), x.stride(1), x.stride(2),
+                     (const long*)nullptr);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor rope_apply_pos(torch::Tensor x, torch::Tensor cs,
+                             torch::Tensor sn, torch::Tensor pos,
+                             bool fwd) {
+  // device-position variant for hipGraph-captured decode
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
+              x.dim() == 4 && x.stride(3) == 1,
+              "rope: x must be 4-D bf16 CUDA with contiguous head dim");
+  TORCH_CHECK(cs.scalar_type() == torch::kFloat &&
+              cs.is_contiguous() && sn.is_contiguous());
+  TORCH_CHECK(pos.is_cuda() && pos.scalar_type() == torch::kLong &&
+              pos.numel() == 1);
+  const int D = x.size(3), S = x.size(2);
+  TORCH_CHECK(D % 16 == 0 && cs.size(1) == D / 2);
+  const long rows = (long)x.size(0) * x.size(1) * S;
+  auto y = torch::empty({x.size(0), x.size(1), x.size(2), x.size(3)},
+                        x.options());
+  const long total = rows * (D / 16);
+  const long nb = (total + BLOCK - 1) / BLOCK;
+  auto stream = at::cuda::getCurrentHIPStream();
+  auto kern = fwd ? rope_kernel<true> : rope_kernel<false>;
+  hipLaunchKernelGGL(kern, dim3((unsigned)std::min(nb, (long)65535 * 8)),
+                     dim3(BLOCK), 0, stream,
+                     (const unsigned short*)x.data_ptr(),
+                     (unsigned short*)y.data_ptr(), cs.data_ptr<float>(),
+                     sn.data_ptr<float>(), rows, S, D, 0,
+                     x.size(1), x.stride(0), x.stride(1), x.stride(2),
+                     (const long*)pos.data_ptr());
   HIP_CHECK_LAST();
   return y;
 }
