@@ -141,6 +141,7 @@ class _GraphedDecoder:
     def step(self) -> torch.Tensor:
         """Emit one token per batch row: (B,) ids.  The returned token is
         the one the PREVIOUS forward predicted; this replay consumes it."""
+        assert self.pos < self.max_seq, "decoder cache is full"
         tok = self.in_tok[:, 0].clone()
         self._graph.replay()
         self.pos += 1
