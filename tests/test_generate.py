@@ -59,3 +59,34 @@ def test_generate_single_token_prompt(gpt2_tiny_model):
     idx = torch.tensor([[5]])
     want = naive_generate(m, idx, 4)
     assert torch.equal(generate(m, idx, 4), want)
+
+
+def test_speculative_equals_greedy(gpt2_tiny_model):
+    """Speculative decode must reproduce the target's greedy output
+    EXACTLY, for any draft."""
+    from torchdistpackage_amd.inference.generate import speculative_generate
+    target = gpt2_tiny_model
+    torch.manual_seed(42)
+    draft = GPT2Model(GPT2Config(vocab_size=512, n_layer=1, n_head=2,
+                                 dim=64, max_seq=64)).eval()
+    torch.manual_seed(5)
+    idx = torch.randint(0, 512, (1, 6))
+    want = generate(target, idx, 12)
+    for k in (1, 3, 4, 7):
+        got = speculative_generate(target, draft, idx, 12, k=k)
+        assert torch.equal(got, want), k
+    # draft == target: every proposal accepted, still exact
+    got = speculative_generate(target, target, idx, 12, k=4)
+    assert torch.equal(got, want)
+
+
+def test_speculative_llama():
+    from torchdistpackage_amd.inference.generate import speculative_generate
+    torch.manual_seed(0)
+    target = LlamaModel(llama_tiny()).eval()
+    torch.manual_seed(9)
+    draft = LlamaModel(llama_tiny()).eval()
+    idx = torch.randint(0, 512, (1, 5))
+    want = generate(target, idx, 10)
+    got = speculative_generate(target, draft, idx, 10, k=3)
+    assert torch.equal(got, want)

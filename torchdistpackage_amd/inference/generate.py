@@ -46,24 +46,31 @@ def _sample(logits: torch.Tensor, greedy: bool, temperature: float,
 
 
 @torch.no_grad()
-def _gpt2_decode_forward(model, idx: torch.Tensor, caches, pos0: int):
+def _gpt2_decode_forward(model, idx: torch.Tensor, caches, pos0: int,
+                         all_logits: bool = False):
     """Forward ``idx`` (B, S_new) through the cached stack; returns the last
-    position's logits (B, V)."""
+    position's logits (B, V), or every position's (B, S_new, V) with
+    ``all_logits`` (speculative verification needs the whole chunk)."""
     B, S = idx.shape
     pos = torch.arange(pos0, pos0 + S, device=idx.device)
     x = model.embed.wte(idx) + model.embed.wpe(pos)[None, :, :]
     x = x.transpose(0, 1).contiguous()          # (S, B, D)
     for blk, (kc, vc) in zip(model.blocks, caches):
         x = blk.decode_step(x, kc, vc, pos0)
+    if all_logits:
+        return model.head(x)                    # (B, S, V)
     return model.head(x[-1:])[:, -1]            # ln_f + tied head, (B, V)
 
 
 @torch.no_grad()
-def _llama_decode_forward(model, idx: torch.Tensor, caches, pos0: int):
+def _llama_decode_forward(model, idx: torch.Tensor, caches, pos0: int,
+                          all_logits: bool = False):
     B, S = idx.shape
     x = model.embed.tok(idx).transpose(0, 1).contiguous()
     for blk, (kc, vc) in zip(model.blocks, caches):
         x = blk.decode_step(x, kc, vc, pos0)
+    if all_logits:
+        return model.head(x)
     return model.head(x[-1:])[:, -1]
 
 
@@ -286,3 +293,64 @@ def generate(model, idx: torch.Tensor, max_new_tokens: int,
         pos0 += chunk.shape[1]
         chunk = nxt[:, None]
     return tokens
+
+
+@torch.no_grad()
+def speculative_generate(target, draft, idx: torch.Tensor,
+                         max_new_tokens: int, k: int = 4) -> torch.Tensor:
+    """Greedy speculative decoding: ``draft`` proposes ``k`` tokens per
+    round, ``target`` verifies the whole chunk in ONE forward.  Output is
+    EXACTLY the target's greedy generation — the draft only changes how
+    many target forwards it takes (verified by the differential test).
+
+    Cache bookkeeping: rejected proposals leave stale entries past the
+    committed length in both caches; they are overwritten sequentially
+    before any later position can attend to them, so a rollback is just a
+    smaller next ``pos0``.  B=1 (speculative decode is per-request —
+    batched rows would accept different lengths each round).
+    """
+    assert get_tp_size() == 1
+    assert idx.shape[0] == 1, "speculative decode is per-request (B=1)"
+    t_fwd = _llama_decode_forward if hasattr(target.embed, "tok") \
+        else _gpt2_decode_forward
+    d_fwd = _llama_decode_forward if hasattr(draft.embed, "tok") \
+        else _gpt2_decode_forward
+    S0 = idx.shape[1]
+    cap = S0 + max_new_tokens + k + 1
+    assert cap <= target.cfg.max_seq and cap <= draft.cfg.max_seq
+    dev = idx.device
+
+    def caches_of(m):
+        cfg = m.cfg
+        return _alloc_caches(cfg.n_layer, 1,
+                             getattr(cfg, "n_kv_head", cfg.n_head), cap,
+                             cfg.dim // cfg.n_head, dev,
+                             next(m.parameters()).dtype)
+
+    t_caches, d_caches = caches_of(target), caches_of(draft)
+    lt = t_fwd(target, idx, t_caches, 0)
+    d_fwd(draft, idx, d_caches, 0)
+    tokens = torch.cat([idx, lt.argmax(-1)[:, None]], dim=1)
+    produced = 1
+    while produced < max_new_tokens:
+        kk = min(k, max_new_tokens - produced)
+        base = tokens.shape[1] - 1        # cache-valid length of both models
+        # ---- draft proposes kk tokens after the last committed token
+        props = []
+        cur = tokens[:, -1:]
+        for j in range(kk):
+            dl = d_fwd(draft, cur, d_caches, base + j)
+            cur = dl.argmax(-1)[:, None]
+            props.append(cur)
+        # ---- target verifies [last_committed, props[:-1]] in one chunk
+        chunk = torch.cat([tokens[:, -1:]] + props[:-1], dim=1)  # (1, kk)
+        tl = t_fwd(target, chunk, t_caches, base, all_logits=True)
+        truth = tl.argmax(-1)             # (1, kk): token after each prefix
+        a = 0
+        while a < kk - 1 and bool(truth[0, a] == props[a][0, 0]):
+            a += 1
+        # accept props[0..a-1] (they matched) plus the target's token at a
+        accepted = props[:a] + [truth[:, a:a + 1]]
+        tokens = torch.cat([tokens] + accepted, dim=1)
+        produced += a + 1
+    return tokens[:, :S0 + max_new_tokens]
